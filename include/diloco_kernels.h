@@ -1,0 +1,150 @@
+/* diloco_kernels.h — C-ABI boundary of the MI355X-native DiLoCo hot path.
+ *
+ * Every entry point here replaces a GPU-math call site of the reference
+ * (PrimeIntellect-ai/OpenDiloco @ 2024-10-08); the reference interface each
+ * one replaces is cited per function.  Conventions:
+ *   - all device pointers are raw, caller-owned, stream-ordered;
+ *   - `stream` is a hipStream_t passed as void*;
+ *   - return value is the hipError_t of the launch (0 == success);
+ *   - no internal synchronisation, no allocation;
+ *   - dk_dtype: 0 = float32, 1 = float16, 2 = bfloat16.
+ *
+ * The Python side (opendiloco_amd/ops.py) binds these through a thin torch
+ * extension (csrc/binding.cpp); a non-torch host could bind them with ctypes
+ * or dlopen directly.
+ */
+#ifndef DILOCO_KERNELS_H
+#define DILOCO_KERNELS_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef void* dkStream; /* hipStream_t */
+
+enum dk_dtype { DK_F32 = 0, DK_F16 = 1, DK_BF16 = 2 };
+
+/* ---- RMSNorm ------------------------------------------------------------
+ * Replaces transformers LlamaRMSNorm fwd/bwd, invoked from the model
+ * forward the reference calls at train_fsdp.py:378 / train_diloco_torch.py:313.
+ * y[r,c] = w[c] * (x[r,c] * invrms[r]);  invrms[r] = rsqrt(mean(x[r,:]^2)+eps)
+ * x,y,w: dtype; invrms: f32[rows] (saved for bwd). */
+int dk_rmsnorm_fwd(void* y, float* invrms, const void* x, const void* w,
+                   int64_t rows, int64_t cols, float eps, int dtype, dkStream stream);
+/* dx: dtype; dw_partial: f32[grid][cols] workspace written deterministically;
+ * grid is returned by dk_rmsnorm_bwd_grid(rows). A second call to
+ * dk_reduce_partials sums dw_partial into dw (f32[cols]). */
+int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy, const void* x,
+                   const void* w, const float* invrms,
+                   int64_t rows, int64_t cols, int grid, int dtype, dkStream stream);
+int dk_rmsnorm_bwd_grid(int64_t rows);
+/* out[c] = sum_g partial[g][c], fixed order (deterministic). out dtype f32. */
+int dk_reduce_partials(float* out, const float* partial, int grid, int64_t cols,
+                       dkStream stream);
+
+/* ---- RoPE ---------------------------------------------------------------
+ * Replaces transformers' apply_rotary_pos_emb (half-split convention:
+ * rotate_half(x) = cat(-x2, x1)) inside the reference's model forward.
+ * x: [n_tok_rows, D] where a row is one (b, h, s) head-vector; costab/sintab:
+ * f32[S, D/2]; row r's position = (r / heads_stride) % S ... simplified:
+ * caller passes seq len S and heads H so pos = (r % (S*?)).  We use layout
+ * [B, H, S, D]: pos = (r % S).  backward=1 applies the transposed rotation. */
+int dk_rope(void* out, const void* x, const float* costab, const float* sintab,
+            int64_t n_rows, int64_t S, int64_t D, int backward, int dtype,
+            dkStream stream);
+
+/* ---- SwiGLU -------------------------------------------------------------
+ * Replaces transformers LlamaMLP's act_fn(gate)*up (silu).
+ * y = silu(gate) * up, elementwise over n elements. */
+int dk_swiglu_fwd(void* y, const void* gate, const void* up, int64_t n,
+                  int dtype, dkStream stream);
+int dk_swiglu_bwd(void* dgate, void* dup, const void* dy, const void* gate,
+                  const void* up, int64_t n, int dtype, dkStream stream);
+
+/* ---- Fused cross-entropy ------------------------------------------------
+ * Replaces the causal-LM loss of LlamaForCausalLM (logits->fp32, shifted CE,
+ * mean over tokens) invoked via model(**batch) at train_fsdp.py:378.
+ * logits: dtype[T, V]; labels: int64[T]; loss_rows/lse: f32[T]. */
+int dk_cross_entropy_fwd(float* loss_rows, float* lse, const void* logits,
+                         const int64_t* labels, int64_t T, int64_t V,
+                         int dtype, dkStream stream);
+/* dlogits = (softmax(logits) - onehot(labels)) * (*dloss);  dloss: f32 device
+ * scalar (upstream grad / T folded in by caller or via scale_per_row). */
+int dk_cross_entropy_bwd(void* dlogits, const void* logits, const float* lse,
+                         const int64_t* labels, const float* dloss, float inv_T,
+                         int64_t T, int64_t V, int dtype, dkStream stream);
+
+/* ---- Flash attention (causal, GQA) --------------------------------------
+ * Replaces torch SDPA inside the reference's model forward (attn_implementation
+ * "sdpa", train_fsdp.py:107).  Layout [B, H, S, D] contiguous; D in {32,64};
+ * o: dtype; lse: f32[B,Hq,S] = m + log(l) saved for bwd.  scale = 1/sqrt(D). */
+int dk_attn_fwd(void* o, float* lse, const void* q, const void* k, const void* v,
+                int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
+                float scale, int dtype, dkStream stream);
+/* delta[b,h,s] = rowsum(do * o), fp32 — preprocessing for bwd. */
+int dk_attn_bwd_preprocess(float* delta, const void* do_, const void* o,
+                           int64_t rows, int64_t D, int dtype, dkStream stream);
+int dk_attn_bwd_dkdv(void* dk_out, void* dv_out, const void* do_, const void* q,
+                     const void* k, const void* v, const float* lse,
+                     const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
+                     int64_t S, int64_t D, float scale, int dtype, dkStream stream);
+int dk_attn_bwd_dq(void* dq_out, const void* do_, const void* q, const void* k,
+                   const void* v, const float* lse, const float* delta,
+                   int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
+                   float scale, int dtype, dkStream stream);
+
+/* ---- Fused AdamW (inner optimizer) --------------------------------------
+ * Replaces the torch AdamW step the reference runs at hivemind_diloco.py:546-550
+ * and train_diloco_torch.py:186,325 (lr cfg, wd=0.1, betas=(0.9,0.95)).
+ * Flat fp32 buffers p, g, m, v of n elements; torch single-tensor op order
+ * (mul_(1-lr*wd); lerp_(m,g,1-b1); v=b2*v+(1-b2)g^2; p -= lr/bc1 * m/(sqrt(v)/sqrt(bc2)+eps)). */
+int dk_fused_adamw(float* p, const float* g, float* m, float* v,
+                   int64_t n, float lr, float beta1, float beta2, float eps,
+                   float weight_decay, int step, dkStream stream);
+
+/* ---- Gradient clipping --------------------------------------------------
+ * Replaces clip_grad_norm_(1.0) at train_fsdp.py:395 / train_diloco_torch.py:323.
+ * Two deterministic passes: partial sums of squares, then finalize+scale.
+ * partials: f32[grid]; grid from dk_gradsq_grid(n). */
+int dk_grad_sq_partials(float* partials, const float* g, int64_t n, int grid,
+                        dkStream stream);
+int dk_gradsq_grid(int64_t n);
+/* total_norm_out (f32 device scalar) = sqrt(sum partials); then
+ * g *= min(1, max_norm / (total_norm + 1e-6))   (torch semantics). */
+int dk_clip_apply(float* g, float* total_norm_out, const float* partials,
+                  int grid, int64_t n, float max_norm, dkStream stream);
+
+/* ---- Outer step (pseudo-gradient + Nesterov SGD + copy-back) ------------
+ * Replaces: pseudo-grad theta_outer - theta_local (hivemind_diloco.py:158-167,
+ * train_diloco_torch.py:342-344). Flat fp32. */
+int dk_pseudo_grad(float* g_out, const float* theta_outer, const float* theta_local,
+                   int64_t n, dkStream stream);
+/* After all-reduce of g: Nesterov SGD on theta_outer (torch SGD math:
+ * buf = mu*buf + g (or buf=g when first=1); d = g + mu*buf;
+ * theta_outer -= lr*d) then theta_local = theta_outer (copy-back,
+ * hivemind_diloco.py:654-665,716-720; train_diloco_torch.py:346-353). */
+int dk_outer_nesterov(float* theta_outer, float* theta_local, float* momentum_buf,
+                      const float* g_avg, int64_t n, float lr, float momentum,
+                      int first_step, dkStream stream);
+
+/* ---- misc ---------------------------------------------------------------
+ * Elementwise scaled cast between f32 and bf16/f16 flat buffers (used for
+ * fp16/bf16 all-reduce payloads when compression is requested). */
+int dk_cast(void* dst, const void* src, int64_t n, int dst_dtype, int src_dtype,
+            dkStream stream);
+
+/* MFMA fragment-layout probes (test-only): writes, for each lane l and slot j,
+ * the (row, k) element index that slot maps to, by multiplying basis matrices.
+ * kind: 0 = A-layout of mfma_f32_16x16x32_bf16, 1 = B-layout, 2 = C-layout. */
+int dk_probe_mfma_16x16x32_bf16(float* out_d, const void* a16x32, const void* b32x16,
+                                dkStream stream);
+
+/* version / build info */
+const char* dk_version(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* DILOCO_KERNELS_H */

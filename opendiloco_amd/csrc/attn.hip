@@ -1,0 +1,687 @@
+// attn.hip — causal flash attention fwd/bwd for gfx950 (CDNA4 MFMA).
+//
+// Replaces torch SDPA inside the reference's Llama forward
+// (attn_implementation "sdpa", reference train_fsdp.py:107; called via
+// model(**batch) at train_fsdp.py:378 / train_diloco_torch.py:313) and its
+// autograd backward (train_fsdp.py:383).
+//
+// Design (MI355X-first, correctness-first structure for round 1):
+//  - mfma_f32_16x16x32_{bf16,f16} tiles; 4 waves (256 thr) per workgroup.
+//  - forward: workgroup owns 64 q rows of one (b, h); waves own 16 q rows
+//    each; K/V staged in LDS by all 256 threads (K row-major [32][D+8] for
+//    direct B-fragment ds_read_b128, V transposed [D][32+8] at staging time
+//    so PV B-fragments are contiguous reads); online softmax in fp32 with
+//    running (m, l) per row, wave-shuffle row reductions (no serial lanes);
+//    P goes through a small per-wave LDS tile to re-shape C-layout ->
+//    A-layout.  lse = m + log(l) saved for backward.
+//  - backward: split into a dK/dV kernel (grid over kv tiles; P^T recomputed
+//    from lse) and a dQ kernel (grid over q tiles) — no atomics anywhere, so
+//    gradients are bit-deterministic run to run (the reference's tests
+//    compare loss traces, test_train.py:82).  GQA handled by writing dK/dV
+//    per q-head; the caller sums the group (ratio == 1 writes directly).
+//
+// MFMA fragment maps used (verified on hardware by dk_probe_mfma_16x16x32,
+// tests/test_gpu_ops.py::test_mfma_probe):
+//   A[16][32]: lane l holds row = l&15,  k = 8*(l>>4) + j   (j = 0..7)
+//   B[32][16]: lane l holds col = l&15,  k = 8*(l>>4) + j
+//   C[16][16]: lane l holds col = l&15,  row = (l>>4)*4 + r (r = 0..3)
+
+#include "dk_common.h"
+#include "../../include/diloco_kernels.h"
+
+#include <math.h>
+
+typedef __attribute__((ext_vector_type(8))) _Float16 halfx8;
+
+template <int DT> struct MFMA16;
+template <> struct MFMA16<2> {
+  using frag = shortx8;
+  static __device__ __forceinline__ floatx4 mma(frag a, frag b, floatx4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct MFMA16<1> {
+  using frag = halfx8;
+  static __device__ __forceinline__ floatx4 mma(frag a, frag b, floatx4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
+
+#define NEG_BIG (-1e30f)
+
+// row-group shuffle reduce: combine over the 16 lanes that share l>>4
+__device__ __forceinline__ float grp16_max(float x) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) x = fmaxf(x, __shfl_xor(x, off, DK_WAVE));
+  return x;
+}
+__device__ __forceinline__ float grp16_sum(float x) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) x += __shfl_xor(x, off, DK_WAVE);
+  return x;
+}
+
+// ======================= forward =======================
+// LDS layout (dynamic): K[32][D+8] | V_T[D][32+8] | P[4][16][32+8]
+template <int DT, int D>
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    typename DTraits<DT>::T* __restrict__ o, float* __restrict__ lse,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    int B, int Hq, int Hkv, int S, float scale) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA16<DT>;
+  using frag = typename MF::frag;
+  constexpr int KT = 32;           // kv tile
+  constexpr int KS = KT + 8;       // padded LDS stride (keys dim)
+  constexpr int DS = D + 8;        // padded LDS stride (channel dim)
+  constexpr int NKC = D / 32;      // mfma K-chunks over channels
+  constexpr int NDN = D / 16;      // output channel tiles
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* K_lds = (T*)smem_raw;                       // [KT][DS]
+  T* VT_lds = K_lds + KT * DS;                   // [D][KS]
+  T* P_lds = VT_lds + D * KS;                    // [4][16][KS]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo = lane & 15;        // col index (key / d-channel)
+  const int hi = lane >> 4;        // k-chunk group & row group
+
+  const int nQT = (S + 63) / 64;
+  int bid = blockIdx.x;
+  const int qt = bid % nQT;
+  const int h = (bid / nQT) % Hq;
+  const int b = bid / (nQT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int q0 = qt * 64 + wave * 16;            // this wave's first q row
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+
+  // Q A-fragments for this wave's 16 rows (row may exceed S on the tail
+  // tile: clamp the load address, rows >= S are masked out of all writes).
+  frag q_frag[NKC];
+  {
+    const int qrow = q0 + lo;
+    const int qr_c = qrow < S ? qrow : S - 1;
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc)
+      q_frag[kc] = *(const frag*)(q + qoff + (int64_t)qr_c * D + kc * 32 + hi * 8);
+  }
+
+  float m_run[4], l_run[4];
+  floatx4 o_acc[NDN];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = NEG_BIG; l_run[r] = 0.f; }
+#pragma unroll
+  for (int dn = 0; dn < NDN; ++dn) o_acc[dn] = (floatx4)(0.f);
+
+  const int kv_end = min(S, qt * 64 + 64);       // causal upper bound
+  const int n_kt = (kv_end + KT - 1) / KT;
+
+  for (int kt = 0; kt < n_kt; ++kt) {
+    const int kbase = kt * KT;
+    __syncthreads();
+    // ---- stage K tile (row-major, padded) + V tile (transposed) ----
+    {
+      constexpr int LPT = (KT * D) / 8;          // 16-B loads for the tile
+      for (int t = threadIdx.x; t < LPT; t += 256) {
+        const int row = t / (D / 8);
+        const int c8 = (t % (D / 8)) * 8;
+        const int krow = kbase + row;
+        const int kr_c = krow < S ? krow : S - 1;
+        frag kv8 = *(const frag*)(k + kvoff + (int64_t)kr_c * D + c8);
+        *(frag*)(K_lds + row * DS + c8) = kv8;
+        frag vv8 = *(const frag*)(v + kvoff + (int64_t)kr_c * D + c8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) VT_lds[(c8 + j) * KS + row] = ((const T*)&vv8)[j];
+      }
+    }
+    __syncthreads();
+
+    // ---- S tile = Q K^T (per wave: 16 q x 32 keys) ----
+    floatx4 sc[2];
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      sc[nt] = (floatx4)(0.f);
+#pragma unroll
+      for (int kc = 0; kc < NKC; ++kc) {
+        frag bk = *(const frag*)(K_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        sc[nt] = MF::mma(q_frag[kc], bk, sc[nt]);
+      }
+    }
+
+    // ---- mask + online softmax ----
+    float p[2][4];
+    float m_new[4], alpha[4], l_add[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + hi * 4 + r;
+      float s0 = sc[0][r] * scale, s1 = sc[1][r] * scale;
+      const int k0 = kbase + lo, k1 = kbase + 16 + lo;
+      if (k0 > qrow || k0 >= S) s0 = NEG_BIG;
+      if (k1 > qrow || k1 >= S) s1 = NEG_BIG;
+      float rm = grp16_max(fmaxf(s0, s1));
+      m_new[r] = fmaxf(m_run[r], rm);
+      alpha[r] = __expf(m_run[r] - m_new[r]);
+      p[0][r] = s0 <= NEG_BIG ? 0.f : __expf(s0 - m_new[r]);
+      p[1][r] = s1 <= NEG_BIG ? 0.f : __expf(s1 - m_new[r]);
+      l_add[r] = grp16_sum(p[0][r] + p[1][r]);
+      l_run[r] = l_run[r] * alpha[r] + l_add[r];
+      m_run[r] = m_new[r];
+    }
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[dn][r] *= alpha[r];
+
+    // ---- P through per-wave LDS: C-layout -> A-layout ----
+    T* Pw = P_lds + wave * 16 * KS;
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) Pw[(hi * 4 + r) * KS + nt * 16 + lo] = TR::fromF(p[nt][r]);
+    // same-wave LDS dependency: compiler inserts lgkmcnt waits; no barrier needed
+    frag pa = *(const frag*)(Pw + lo * KS + hi * 8);
+
+    // ---- O += P V ----
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn) {
+      frag bv = *(const frag*)(VT_lds + (dn * 16 + lo) * KS + hi * 8);
+      o_acc[dn] = MF::mma(pa, bv, o_acc[dn]);
+    }
+  }
+
+  // ---- epilogue ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + hi * 4 + r;
+    if (qrow >= S) continue;
+    const float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn)
+      o[qoff + (int64_t)qrow * D + dn * 16 + lo] = TR::fromF(o_acc[dn][r] * inv_l);
+    if (lo == 0) lse[((int64_t)b * Hq + h) * S + qrow] = m_run[r] + __logf(l_run[r]);
+  }
+}
+
+// ======================= bwd preprocess: delta = rowsum(do*o) =======================
+template <int DT>
+__global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
+                                    const typename DTraits<DT>::T* __restrict__ do_,
+                                    const typename DTraits<DT>::T* __restrict__ o,
+                                    int64_t rows, int D) {
+  using TR = DTraits<DT>;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t w0 = (int64_t)blockIdx.x * 4 + wave;
+  const int64_t stride = (int64_t)gridDim.x * 4;
+  for (int64_t r = w0; r < rows; r += stride) {
+    float s = 0.f;
+    for (int c = lane; c < D; c += 64)
+      s += TR::toF(do_[r * D + c]) * TR::toF(o[r * D + c]);
+    s = wave_reduce_sum(s);
+    if (lane == 0) delta[r] = s;
+  }
+}
+
+// ======================= bwd dK/dV =======================
+// grid over (b, hq, kv-tile of 64 keys); wave owns 16 keys.  Loop q tiles of
+// 32.  LDS: Q[32][D+8] | Q_T[D][32+8] | dO[32][D+8] | dO_T[D][32+8] |
+//          lse[32] f32 | delta[32] f32 | P_T[4][16][32+8]
+template <int DT, int D>
+__global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
+    typename DTraits<DT>::T* __restrict__ dk_out,
+    typename DTraits<DT>::T* __restrict__ dv_out,
+    const typename DTraits<DT>::T* __restrict__ do_,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    int B, int Hq, int Hkv, int S, float scale) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA16<DT>;
+  using frag = typename MF::frag;
+  constexpr int QT = 32;
+  constexpr int QS = QT + 8;
+  constexpr int DS = D + 8;
+  constexpr int NKC = D / 32;
+  constexpr int NDN = D / 16;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* Q_lds = (T*)smem_raw;                   // [QT][DS]
+  T* QT_lds = Q_lds + QT * DS;               // [D][QS]
+  T* dO_lds = QT_lds + D * QS;               // [QT][DS]
+  T* dOT_lds = dO_lds + QT * DS;             // [D][QS]
+  T* PT_lds = dOT_lds + D * QS;              // [4][16][QS]  (P^T tiles)
+  T* DS_lds = PT_lds + 4 * 16 * QS;          // [4][16][QS]  (dS^T tiles; separate
+                                             //  buffer: avoids an LDS WAR hazard
+                                             //  between the P^T A-frag read and
+                                             //  the dS^T writes in one iteration)
+  float* lse_lds = (float*)(DS_lds + 4 * 16 * QS);  // [QT]
+  float* dl_lds = lse_lds + QT;                     // [QT]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo = lane & 15;
+  const int hi = lane >> 4;
+
+  const int nKT = (S + 63) / 64;
+  int bid = blockIdx.x;
+  const int kt = bid % nKT;
+  const int h = (bid / nKT) % Hq;
+  const int b = bid / (nKT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int k0 = kt * 64 + wave * 16;        // this wave's first key
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+
+  // K,V A-fragments for this wave's 16 keys
+  frag k_frag[NKC], v_frag[NKC];
+  {
+    const int krow = k0 + lo;
+    const int kr_c = krow < S ? krow : S - 1;
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      k_frag[kc] = *(const frag*)(k + kvoff + (int64_t)kr_c * D + kc * 32 + hi * 8);
+      v_frag[kc] = *(const frag*)(v + kvoff + (int64_t)kr_c * D + kc * 32 + hi * 8);
+    }
+  }
+
+  floatx4 dv_acc[NDN], dk_acc[NDN];
+#pragma unroll
+  for (int dn = 0; dn < NDN; ++dn) { dv_acc[dn] = (floatx4)(0.f); dk_acc[dn] = (floatx4)(0.f); }
+
+  const int qstart = (kt * 64) / QT;         // first q tile that sees these keys
+  const int nQT2 = (S + QT - 1) / QT;
+
+  for (int qt = qstart; qt < nQT2; ++qt) {
+    const int qbase = qt * QT;
+    __syncthreads();
+    {  // stage Q, dO (+ transposes), lse, delta
+      constexpr int LPT = (QT * D) / 8;
+      for (int t = threadIdx.x; t < LPT; t += 256) {
+        const int row = t / (D / 8);
+        const int c8 = (t % (D / 8)) * 8;
+        const int qrow = qbase + row;
+        const int qr_c = qrow < S ? qrow : S - 1;
+        frag q8 = *(const frag*)(q + qoff + (int64_t)qr_c * D + c8);
+        *(frag*)(Q_lds + row * DS + c8) = q8;
+        frag d8 = *(const frag*)(do_ + qoff + (int64_t)qr_c * D + c8);
+        *(frag*)(dO_lds + row * DS + c8) = d8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          QT_lds[(c8 + j) * QS + row] = ((const T*)&q8)[j];
+          dOT_lds[(c8 + j) * QS + row] = ((const T*)&d8)[j];
+        }
+      }
+      for (int t = threadIdx.x; t < QT; t += 256) {
+        const int qrow = qbase + t;
+        const int qr_c = qrow < S ? qrow : S - 1;
+        lse_lds[t] = lse[lseoff + qr_c];
+        dl_lds[t] = delta[lseoff + qr_c];
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T = K Q^T (16 keys x 32 q), P^T = exp(scale*S^T - lse) ----
+    float pt[2][4], dst[2][4];
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      floatx4 st = (floatx4)(0.f);
+#pragma unroll
+      for (int kc = 0; kc < NKC; ++kc) {
+        frag bq = *(const frag*)(Q_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        st = MF::mma(k_frag[kc], bq, st);
+      }
+      // dP^T = V dO^T
+      floatx4 dpt = (floatx4)(0.f);
+#pragma unroll
+      for (int kc = 0; kc < NKC; ++kc) {
+        frag bd = *(const frag*)(dO_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        dpt = MF::mma(v_frag[kc], bd, dpt);
+      }
+      const int qcol = qbase + nt * 16 + lo;
+      const float lse_q = lse_lds[nt * 16 + lo];
+      const float dl_q = dl_lds[nt * 16 + lo];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int krow = k0 + hi * 4 + r;
+        float pv = 0.f;
+        if (krow <= qcol && krow < S && qcol < S)
+          pv = __expf(st[r] * scale - lse_q);
+        pt[nt][r] = pv;
+        dst[nt][r] = pv * (dpt[r] - dl_q) * scale;
+      }
+    }
+
+    // ---- dV += P^T dO  (A = P^T via LDS; B = dO_T) ----
+    T* Pw = PT_lds + wave * 16 * QS;
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) Pw[(hi * 4 + r) * QS + nt * 16 + lo] = TR::fromF(pt[nt][r]);
+    frag pa = *(const frag*)(Pw + lo * QS + hi * 8);
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn) {
+      frag bd = *(const frag*)(dOT_lds + (dn * 16 + lo) * QS + hi * 8);
+      dv_acc[dn] = MF::mma(pa, bd, dv_acc[dn]);
+    }
+
+    // ---- dK += dS^T Q  (A = dS^T via LDS; B = Q_T) ----
+    T* Dw = DS_lds + wave * 16 * QS;
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) Dw[(hi * 4 + r) * QS + nt * 16 + lo] = TR::fromF(dst[nt][r]);
+    frag da = *(const frag*)(Dw + lo * QS + hi * 8);
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn) {
+      frag bq = *(const frag*)(QT_lds + (dn * 16 + lo) * QS + hi * 8);
+      dk_acc[dn] = MF::mma(da, bq, dk_acc[dn]);
+    }
+  }
+
+  // ---- write dK, dV (per q-head layout [B,Hq,S,D]; caller sums GQA groups) ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int krow = k0 + hi * 4 + r;
+    if (krow >= S) continue;
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn) {
+      dk_out[qoff + (int64_t)krow * D + dn * 16 + lo] = TR::fromF(dk_acc[dn][r]);
+      dv_out[qoff + (int64_t)krow * D + dn * 16 + lo] = TR::fromF(dv_acc[dn][r]);
+    }
+  }
+}
+
+// ======================= bwd dQ =======================
+// grid over (b, hq, q-tile of 64); wave owns 16 q rows.  Loop kv tiles of 32.
+// LDS: K[32][D+8] | V[32][D+8] | K_T[D][32+8] | dS[4][16][32+8]
+template <int DT, int D>
+__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+    typename DTraits<DT>::T* __restrict__ dq_out,
+    const typename DTraits<DT>::T* __restrict__ do_,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    int B, int Hq, int Hkv, int S, float scale) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA16<DT>;
+  using frag = typename MF::frag;
+  constexpr int KT = 32;
+  constexpr int KS = KT + 8;
+  constexpr int DS = D + 8;
+  constexpr int NKC = D / 32;
+  constexpr int NDN = D / 16;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* K_lds = (T*)smem_raw;                    // [KT][DS]
+  T* V_lds = K_lds + KT * DS;                 // [KT][DS]
+  T* KT_lds = V_lds + KT * DS;                // [D][KS]
+  T* S_lds = KT_lds + D * KS;                 // [4][16][KS]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo = lane & 15;
+  const int hi = lane >> 4;
+
+  const int nQT = (S + 63) / 64;
+  int bid = blockIdx.x;
+  const int qt = bid % nQT;
+  const int h = (bid / nQT) % Hq;
+  const int b = bid / (nQT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int q0 = qt * 64 + wave * 16;
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+
+  frag q_frag[NKC], do_frag[NKC];
+  float lse_r[4], dl_r[4];
+  {
+    const int qrow = q0 + lo;
+    const int qr_c = qrow < S ? qrow : S - 1;
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      q_frag[kc] = *(const frag*)(q + qoff + (int64_t)qr_c * D + kc * 32 + hi * 8);
+      do_frag[kc] = *(const frag*)(do_ + qoff + (int64_t)qr_c * D + kc * 32 + hi * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int rr = q0 + hi * 4 + r;
+      const int rr_c = rr < S ? rr : S - 1;
+      lse_r[r] = lse[lseoff + rr_c];
+      dl_r[r] = delta[lseoff + rr_c];
+    }
+  }
+
+  floatx4 dq_acc[NDN];
+#pragma unroll
+  for (int dn = 0; dn < NDN; ++dn) dq_acc[dn] = (floatx4)(0.f);
+
+  const int kv_end = min(S, qt * 64 + 64);
+  const int n_kt = (kv_end + KT - 1) / KT;
+
+  for (int kt = 0; kt < n_kt; ++kt) {
+    const int kbase = kt * KT;
+    __syncthreads();
+    {
+      constexpr int LPT = (KT * D) / 8;
+      for (int t = threadIdx.x; t < LPT; t += 256) {
+        const int row = t / (D / 8);
+        const int c8 = (t % (D / 8)) * 8;
+        const int krow = kbase + row;
+        const int kr_c = krow < S ? krow : S - 1;
+        frag k8 = *(const frag*)(k + kvoff + (int64_t)kr_c * D + c8);
+        *(frag*)(K_lds + row * DS + c8) = k8;
+        frag v8 = *(const frag*)(v + kvoff + (int64_t)kr_c * D + c8);
+        *(frag*)(V_lds + row * DS + c8) = v8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) KT_lds[(c8 + j) * KS + row] = ((const T*)&k8)[j];
+      }
+    }
+    __syncthreads();
+
+    float ds[2][4];
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      floatx4 st = (floatx4)(0.f), dpt = (floatx4)(0.f);
+#pragma unroll
+      for (int kc = 0; kc < NKC; ++kc) {
+        frag bk = *(const frag*)(K_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        st = MF::mma(q_frag[kc], bk, st);
+        frag bv = *(const frag*)(V_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        dpt = MF::mma(do_frag[kc], bv, dpt);
+      }
+      const int kcol = kbase + nt * 16 + lo;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + hi * 4 + r;
+        float pv = 0.f;
+        if (kcol <= qrow && kcol < S && qrow < S)
+          pv = __expf(st[r] * scale - lse_r[r]);
+        ds[nt][r] = pv * (dpt[r] - dl_r[r]) * scale;
+      }
+    }
+
+    // dQ += dS K  (A = dS via LDS; B = K_T)
+    T* Sw = S_lds + wave * 16 * KS;
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) Sw[(hi * 4 + r) * KS + nt * 16 + lo] = TR::fromF(ds[nt][r]);
+    frag da = *(const frag*)(Sw + lo * KS + hi * 8);
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn) {
+      frag bk = *(const frag*)(KT_lds + (dn * 16 + lo) * KS + hi * 8);
+      dq_acc[dn] = MF::mma(da, bk, dq_acc[dn]);
+    }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + hi * 4 + r;
+    if (qrow >= S) continue;
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn)
+      dq_out[qoff + (int64_t)qrow * D + dn * 16 + lo] = TR::fromF(dq_acc[dn][r]);
+  }
+}
+
+// ======================= C-ABI wrappers =======================
+
+template <int DT, int D>
+static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
+                           const void* v, int64_t B, int64_t Hq, int64_t Hkv,
+                           int64_t S, float scale, dkStream stream) {
+  using T = typename DTraits<DT>::T;
+  const int nQT = (int)((S + 63) / 64);
+  const int grid = (int)(B * Hq * nQT);
+  constexpr int KT = 32, KS = KT + 8, DS = D + 8;
+  const size_t lds = sizeof(T) * (KT * DS + D * KS + 4 * 16 * KS);
+  hipLaunchKernelGGL((attn_fwd_kernel<DT, D>), dim3(grid), dim3(256), lds,
+                     (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
+                     (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale);
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int dk_attn_fwd(void* o, float* lse, const void* q, const void* k,
+                           const void* v, int64_t B, int64_t Hq, int64_t Hkv,
+                           int64_t S, int64_t D, float scale, int dtype,
+                           dkStream stream) {
+  if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
+  if (D == 64) {
+    if (dtype == 2) return launch_attn_fwd<2, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
+    return launch_attn_fwd<1, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
+  } else if (D == 32) {
+    if (dtype == 2) return launch_attn_fwd<2, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
+    return launch_attn_fwd<1, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
+  }
+  return (int)hipErrorInvalidValue;
+}
+
+extern "C" int dk_attn_bwd_preprocess(float* delta, const void* do_, const void* o,
+                                      int64_t rows, int64_t D, int dtype,
+                                      dkStream stream) {
+  int64_t g = (rows + 3) / 4;  // 4 waves per block, one row per wave-iteration
+  int grid = (int)(g > 2048 ? 2048 : (g < 1 ? 1 : g));
+  DK_DISPATCH_DT(dtype, {
+    if constexpr (kDT != 0) {
+      using T = typename DTraits<kDT>::T;
+      hipLaunchKernelGGL((attn_bwd_pre_kernel<kDT>), dim3(grid), dim3(256), 0,
+                         (hipStream_t)stream, delta, (const T*)do_, (const T*)o,
+                         rows, (int)D);
+    } else {
+      return (int)hipErrorInvalidValue;
+    }
+  });
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
+template <int DT, int D>
+static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const void* q,
+                                const void* k, const void* v, const float* lse,
+                                const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
+                                int64_t S, float scale, dkStream stream) {
+  using T = typename DTraits<DT>::T;
+  const int nKT = (int)((S + 63) / 64);
+  const int grid = (int)(B * Hq * nKT);
+  constexpr int QT = 32, QS = QT + 8, DS = D + 8;
+  const size_t lds = sizeof(T) * (2 * QT * DS + 2 * D * QS + 2 * 4 * 16 * QS) + sizeof(float) * 2 * QT;
+  hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(256), lds,
+                     (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
+                     (const T*)q, (const T*)k, (const T*)v, lse, delta,
+                     (int)B, (int)Hq, (int)Hkv, (int)S, scale);
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int dk_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const void* q,
+                                const void* k, const void* v, const float* lse,
+                                const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
+                                int64_t S, int64_t D, float scale, int dtype,
+                                dkStream stream) {
+  if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
+  if (D == 64) {
+    if (dtype == 2) return launch_attn_bwd_dkdv<2, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    return launch_attn_bwd_dkdv<1, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+  } else if (D == 32) {
+    if (dtype == 2) return launch_attn_bwd_dkdv<2, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    return launch_attn_bwd_dkdv<1, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+  }
+  return (int)hipErrorInvalidValue;
+}
+
+template <int DT, int D>
+static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const void* k,
+                              const void* v, const float* lse, const float* delta,
+                              int64_t B, int64_t Hq, int64_t Hkv, int64_t S,
+                              float scale, dkStream stream) {
+  using T = typename DTraits<DT>::T;
+  const int nQT = (int)((S + 63) / 64);
+  const int grid = (int)(B * Hq * nQT);
+  constexpr int KT = 32, KS = KT + 8, DS = D + 8;
+  const size_t lds = sizeof(T) * (2 * KT * DS + D * KS + 4 * 16 * KS);
+  hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(256), lds,
+                     (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
+                     (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
+                     (int)Hkv, (int)S, scale);
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int dk_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const void* k,
+                              const void* v, const float* lse, const float* delta,
+                              int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
+                              float scale, int dtype, dkStream stream) {
+  if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
+  if (D == 64) {
+    if (dtype == 2) return launch_attn_bwd_dq<2, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    return launch_attn_bwd_dq<1, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+  } else if (D == 32) {
+    if (dtype == 2) return launch_attn_bwd_dq<2, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    return launch_attn_bwd_dq<1, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+  }
+  return (int)hipErrorInvalidValue;
+}
+
+// ======================= MFMA layout probe (test-only) =======================
+// D = A[16][32] x B[32][16] using the fragment maps above; out is [16][16]
+// f32 row-major.  A/B given row-major bf16.  The GPU test feeds asymmetric
+// random matrices and compares with a host matmul (guide §5.4 rule 16).
+__global__ void probe_mfma_kernel(float* __restrict__ out,
+                                  const unsigned short* __restrict__ a,
+                                  const unsigned short* __restrict__ b) {
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 15, hi = lane >> 4;
+  shortx8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ((unsigned short*)&af)[j] = a[lo * 32 + hi * 8 + j];        // A[row=lo][k=8*hi+j]
+    ((unsigned short*)&bf)[j] = b[(hi * 8 + j) * 16 + lo];      // B[k=8*hi+j][col=lo]
+  }
+  floatx4 c = (floatx4)(0.f);
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) out[(hi * 4 + r) * 16 + lo] = c[r];  // C[row=(hi*4+r)][col=lo]
+}
+
+extern "C" int dk_probe_mfma_16x16x32_bf16(float* out_d, const void* a16x32,
+                                           const void* b32x16, dkStream stream) {
+  hipLaunchKernelGGL(probe_mfma_kernel, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                     out_d, (const unsigned short*)a16x32, (const unsigned short*)b32x16);
+  DK_CHECK_LAUNCH();
+  return 0;
+}

@@ -1,0 +1,243 @@
+// binding.cpp — thin torch extension over the C-ABI kernel library.
+//
+// This file is host-only glue: tensor validation, output allocation (via the
+// torch caching allocator), dtype mapping and stream plumbing.  All device
+// math lives behind include/diloco_kernels.h (libdiloco_kernels.so).
+// Written directly against torch's native HIP/ROCm surface (c10::hip) —
+// no CUDA-compat layer.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "../../include/diloco_kernels.h"
+
+namespace {
+
+int dt_of(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return DK_F32;
+    case at::kHalf: return DK_F16;
+    case at::kBFloat16: return DK_BF16;
+    default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+void* stream() {
+  return (void*)c10::hip::getCurrentHIPStream().stream();
+}
+
+#define DK_OK(call)                                                        \
+  do {                                                                     \
+    int rc_ = (call);                                                      \
+    TORCH_CHECK(rc_ == 0, #call, " failed with hip error code ", rc_);     \
+  } while (0)
+
+#define CHECK_DEV_CONTIG(t) \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be a contiguous device tensor")
+
+// ---- RMSNorm ----
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps) {
+  CHECK_DEV_CONTIG(x);
+  CHECK_DEV_CONTIG(w);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  DK_OK(dk_rmsnorm_fwd(y.data_ptr(), invrms.data_ptr<float>(), x.data_ptr(), w.data_ptr(),
+                       rows, cols, (float)eps, dt_of(x), stream()));
+  return {y, invrms};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                    const at::Tensor& w, const at::Tensor& invrms) {
+  CHECK_DEV_CONTIG(dy);
+  CHECK_DEV_CONTIG(x);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  const int grid = dk_rmsnorm_bwd_grid(rows);
+  auto dx = at::empty_like(x);
+  auto dwp = at::zeros({grid, cols}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({cols}, x.options().dtype(at::kFloat));
+  DK_OK(dk_rmsnorm_bwd(dx.data_ptr(), dwp.data_ptr<float>(), dy.data_ptr(), x.data_ptr(),
+                       w.data_ptr(), invrms.data_ptr<float>(), rows, cols, grid,
+                       dt_of(x), stream()));
+  DK_OK(dk_reduce_partials(dw.data_ptr<float>(), dwp.data_ptr<float>(), grid, cols, stream()));
+  return {dx, dw};
+}
+
+// ---- RoPE ----
+at::Tensor rope(const at::Tensor& x, const at::Tensor& costab, const at::Tensor& sintab,
+                int64_t S, bool backward) {
+  CHECK_DEV_CONTIG(x);
+  const int64_t D = x.size(-1);
+  const int64_t n_rows = x.numel() / D;
+  auto out = at::empty_like(x);
+  DK_OK(dk_rope(out.data_ptr(), x.data_ptr(), costab.data_ptr<float>(),
+                sintab.data_ptr<float>(), n_rows, S, D, backward ? 1 : 0, dt_of(x), stream()));
+  return out;
+}
+
+// ---- SwiGLU ----
+at::Tensor swiglu_fwd(const at::Tensor& gate, const at::Tensor& up) {
+  CHECK_DEV_CONTIG(gate);
+  CHECK_DEV_CONTIG(up);
+  auto y = at::empty_like(gate);
+  DK_OK(dk_swiglu_fwd(y.data_ptr(), gate.data_ptr(), up.data_ptr(), gate.numel(),
+                      dt_of(gate), stream()));
+  return y;
+}
+
+std::vector<at::Tensor> swiglu_bwd(const at::Tensor& dy, const at::Tensor& gate,
+                                   const at::Tensor& up) {
+  auto dgate = at::empty_like(gate);
+  auto dup = at::empty_like(up);
+  DK_OK(dk_swiglu_bwd(dgate.data_ptr(), dup.data_ptr(), dy.data_ptr(), gate.data_ptr(),
+                      up.data_ptr(), gate.numel(), dt_of(gate), stream()));
+  return {dgate, dup};
+}
+
+// ---- cross entropy ----
+std::vector<at::Tensor> ce_fwd(const at::Tensor& logits, const at::Tensor& labels) {
+  CHECK_DEV_CONTIG(logits);
+  const int64_t V = logits.size(-1);
+  const int64_t T = logits.numel() / V;
+  auto loss_rows = at::empty({T}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({T}, logits.options().dtype(at::kFloat));
+  DK_OK(dk_cross_entropy_fwd(loss_rows.data_ptr<float>(), lse.data_ptr<float>(),
+                             logits.data_ptr(), labels.data_ptr<int64_t>(), T, V,
+                             dt_of(logits), stream()));
+  return {loss_rows, lse};
+}
+
+at::Tensor ce_bwd(const at::Tensor& logits, const at::Tensor& lse, const at::Tensor& labels,
+                  const at::Tensor& dloss, double inv_T) {
+  auto dlogits = at::empty_like(logits);
+  const int64_t V = logits.size(-1);
+  const int64_t T = logits.numel() / V;
+  DK_OK(dk_cross_entropy_bwd(dlogits.data_ptr(), logits.data_ptr(), lse.data_ptr<float>(),
+                             labels.data_ptr<int64_t>(), dloss.data_ptr<float>(),
+                             (float)inv_T, T, V, dt_of(logits), stream()));
+  return dlogits;
+}
+
+// ---- attention ----
+std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                                 const at::Tensor& v, double scale) {
+  CHECK_DEV_CONTIG(q);
+  CHECK_DEV_CONTIG(k);
+  CHECK_DEV_CONTIG(v);
+  const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
+  const int64_t Hkv = k.size(1);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  DK_OK(dk_attn_fwd(o.data_ptr(), lse.data_ptr<float>(), q.data_ptr(), k.data_ptr(),
+                    v.data_ptr(), B, Hq, Hkv, S, D, (float)scale, dt_of(q), stream()));
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd(const at::Tensor& do_, const at::Tensor& q,
+                                 const at::Tensor& k, const at::Tensor& v,
+                                 const at::Tensor& o, const at::Tensor& lse, double scale) {
+  CHECK_DEV_CONTIG(do_);
+  const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
+  const int64_t Hkv = k.size(1);
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  DK_OK(dk_attn_bwd_preprocess(delta.data_ptr<float>(), do_.data_ptr(), o.data_ptr(),
+                               B * Hq * S, D, dt_of(q), stream()));
+  auto dq = at::empty_like(q);
+  // dk/dv computed per q-head; python caller sums GQA groups when Hq != Hkv
+  auto dk_full = at::empty({B, Hq, S, D}, q.options());
+  auto dv_full = at::empty({B, Hq, S, D}, q.options());
+  DK_OK(dk_attn_bwd_dkdv(dk_full.data_ptr(), dv_full.data_ptr(), do_.data_ptr(),
+                         q.data_ptr(), k.data_ptr(), v.data_ptr(), lse.data_ptr<float>(),
+                         delta.data_ptr<float>(), B, Hq, Hkv, S, D, (float)scale,
+                         dt_of(q), stream()));
+  DK_OK(dk_attn_bwd_dq(dq.data_ptr(), do_.data_ptr(), q.data_ptr(), k.data_ptr(),
+                       v.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       B, Hq, Hkv, S, D, (float)scale, dt_of(q), stream()));
+  return {dq, dk_full, dv_full};
+}
+
+// ---- optimizer / outer step ----
+void fused_adamw(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
+                 double lr, double beta1, double beta2, double eps, double weight_decay,
+                 int64_t step) {
+  CHECK_DEV_CONTIG(p);
+  DK_OK(dk_fused_adamw(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), p.numel(), (float)lr, (float)beta1,
+                       (float)beta2, (float)eps, (float)weight_decay, (int)step, stream()));
+}
+
+at::Tensor clip_grad_(at::Tensor& g, double max_norm) {
+  CHECK_DEV_CONTIG(g);
+  const int grid = dk_gradsq_grid(g.numel());
+  auto partials = at::empty({grid}, g.options().dtype(at::kFloat));
+  auto out2 = at::empty({2}, g.options().dtype(at::kFloat));
+  DK_OK(dk_grad_sq_partials(partials.data_ptr<float>(), g.data_ptr<float>(), g.numel(),
+                            grid, stream()));
+  DK_OK(dk_clip_apply(g.data_ptr<float>(), out2.data_ptr<float>(), partials.data_ptr<float>(),
+                      grid, g.numel(), (float)max_norm, stream()));
+  return out2;  // [total_norm, applied_coef]
+}
+
+at::Tensor grad_norm(const at::Tensor& g) {
+  // norm only (no scaling): reuse partials + finalize with max_norm=inf
+  const int grid = dk_gradsq_grid(g.numel());
+  auto partials = at::empty({grid}, g.options().dtype(at::kFloat));
+  DK_OK(dk_grad_sq_partials(partials.data_ptr<float>(), g.data_ptr<float>(), g.numel(),
+                            grid, stream()));
+  return partials.sum().sqrt();
+}
+
+void pseudo_grad(at::Tensor& g_out, const at::Tensor& theta_outer,
+                 const at::Tensor& theta_local) {
+  DK_OK(dk_pseudo_grad(g_out.data_ptr<float>(), theta_outer.data_ptr<float>(),
+                       theta_local.data_ptr<float>(), g_out.numel(), stream()));
+}
+
+void outer_nesterov(at::Tensor& theta_outer, at::Tensor& theta_local, at::Tensor& buf,
+                    const at::Tensor& g_avg, double lr, double momentum, bool first) {
+  DK_OK(dk_outer_nesterov(theta_outer.data_ptr<float>(), theta_local.data_ptr<float>(),
+                          buf.data_ptr<float>(), g_avg.data_ptr<float>(),
+                          theta_outer.numel(), (float)lr, (float)momentum, first ? 1 : 0,
+                          stream()));
+}
+
+void cast_(at::Tensor& dst, const at::Tensor& src) {
+  TORCH_CHECK(dst.numel() == src.numel());
+  DK_OK(dk_cast(dst.data_ptr(), src.data_ptr(), dst.numel(), dt_of(dst), dt_of(src), stream()));
+}
+
+// ---- probe (test-only) ----
+at::Tensor probe_mfma(const at::Tensor& a, const at::Tensor& b) {
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 && a.numel() == 16 * 32);
+  TORCH_CHECK(b.scalar_type() == at::kBFloat16 && b.numel() == 32 * 16);
+  auto out = at::zeros({16, 16}, a.options().dtype(at::kFloat));
+  DK_OK(dk_probe_mfma_16x16x32_bf16(out.data_ptr<float>(), a.data_ptr(), b.data_ptr(),
+                                    stream()));
+  return out;
+}
+
+std::string version() { return dk_version(); }
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope", &rope);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("fused_adamw", &fused_adamw);
+  m.def("clip_grad_", &clip_grad_);
+  m.def("grad_norm", &grad_norm);
+  m.def("pseudo_grad", &pseudo_grad);
+  m.def("outer_nesterov", &outer_nesterov);
+  m.def("cast_", &cast_);
+  m.def("probe_mfma", &probe_mfma);
+  m.def("version", &version);
+}

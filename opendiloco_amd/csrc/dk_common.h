@@ -1,0 +1,124 @@
+// dk_common.h — shared device helpers for the MI355X (gfx950) DiLoCo kernels.
+// CDNA4: wavefront = 64 lanes; block sizes are multiples of 64.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define DK_WAVE 64
+
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+typedef __attribute__((ext_vector_type(8))) short shortx8;
+typedef __attribute__((ext_vector_type(4))) short shortx4;
+
+// ---------- scalar dtype conversions ----------
+__device__ __forceinline__ float bf16_to_f32(unsigned short u) {
+  union { unsigned int i; float f; } c;
+  c.i = ((unsigned int)u) << 16;
+  return c.f;
+}
+
+__device__ __forceinline__ unsigned short f32_to_bf16(float f) {
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  if ((c.i & 0x7fffffffu) > 0x7f800000u) return 0x7fc0u;  // NaN
+  unsigned int lsb = (c.i >> 16) & 1u;
+  return (unsigned short)((c.i + 0x7fffu + lsb) >> 16);   // round-to-nearest-even
+}
+
+// ---------- dtype traits: DT 0=f32, 1=f16, 2=bf16 ----------
+template <int DT> struct DTraits;
+
+template <> struct DTraits<0> {
+  using T = float;
+  static __device__ __forceinline__ float toF(T x) { return x; }
+  static __device__ __forceinline__ T fromF(float x) { return x; }
+};
+template <> struct DTraits<1> {
+  using T = _Float16;
+  static __device__ __forceinline__ float toF(T x) { return (float)x; }
+  static __device__ __forceinline__ T fromF(float x) { return (_Float16)x; }
+};
+template <> struct DTraits<2> {
+  using T = unsigned short;  // bf16 bit pattern
+  static __device__ __forceinline__ float toF(T x) { return bf16_to_f32(x); }
+  static __device__ __forceinline__ T fromF(float x) { return f32_to_bf16(x); }
+};
+
+// vector-of-8 load/store for 16-bit types (16 B — the coalescing sweet spot,
+// cdna_hip_programming.md G13), vector-of-4 for f32.
+template <typename T> struct VecIO;
+template <> struct VecIO<float> {
+  static constexpr int W = 4;
+  using V = floatx4;
+};
+template <> struct VecIO<_Float16> {
+  static constexpr int W = 8;
+  using V = shortx8;
+};
+template <> struct VecIO<unsigned short> {
+  static constexpr int W = 8;
+  using V = shortx8;
+};
+
+// ---------- wave + block reductions ----------
+__device__ __forceinline__ float wave_reduce_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, DK_WAVE);
+  return x;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, DK_WAVE));
+  return x;
+}
+
+// block reduce over NT threads (NT multiple of 64, <= 1024); smem: float[NT/64]
+template <int NT>
+__device__ __forceinline__ float block_reduce_sum(float x, float* smem) {
+  constexpr int NW = NT / DK_WAVE;
+  const int wid = threadIdx.x / DK_WAVE;
+  x = wave_reduce_sum(x);
+  if ((threadIdx.x & (DK_WAVE - 1)) == 0) smem[wid] = x;
+  __syncthreads();
+  float r = 0.f;
+#pragma unroll
+  for (int i = 0; i < NW; ++i) r += smem[i];
+  __syncthreads();
+  return r;
+}
+
+template <int NT>
+__device__ __forceinline__ float block_reduce_max(float x, float* smem) {
+  constexpr int NW = NT / DK_WAVE;
+  const int wid = threadIdx.x / DK_WAVE;
+  x = wave_reduce_max(x);
+  if ((threadIdx.x & (DK_WAVE - 1)) == 0) smem[wid] = x;
+  __syncthreads();
+  float r = -INFINITY;
+#pragma unroll
+  for (int i = 0; i < NW; ++i) r = fmaxf(r, smem[i]);
+  __syncthreads();
+  return r;
+}
+
+#define DK_CHECK_LAUNCH() \
+  do { hipError_t e_ = hipGetLastError(); if (e_ != hipSuccess) return (int)e_; } while (0)
+
+// dispatch over dtype (0=f32,1=f16,2=bf16)
+#define DK_DISPATCH_DT(dtype, ...)                    \
+  switch (dtype) {                                    \
+    case 0: { constexpr int kDT = 0; __VA_ARGS__; break; } \
+    case 1: { constexpr int kDT = 1; __VA_ARGS__; break; } \
+    case 2: { constexpr int kDT = 2; __VA_ARGS__; break; } \
+    default: return (int)hipErrorInvalidValue;        \
+  }
+
+// grid sizing for memory-bound grid-stride kernels (G11: cap ~2048 blocks)
+static inline int dk_stream_grid(int64_t work_items, int block) {
+  int64_t g = (work_items + block - 1) / block;
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  return (int)g;
+}
