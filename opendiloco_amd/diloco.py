@@ -1,0 +1,356 @@
+"""DiLoCoOptimizer — MI355X-native drop-in for the reference's
+``open_diloco.hivemind_diloco.DiLoCoOptimizer`` (hivemind_diloco.py:303-738).
+
+Same constructor signature and attribute surface (the ones the reference CLI
+and tests actually use: .step(scaler=), .inner_optimizer,
+.state_averager.optimizer, .param_groups, .state_dict()/.load_state_dict(),
+.tracker.pause_updates(), .tracker.global_progress.num_peers, .local_epoch,
+.load_state_from_peers(), .diloco_grad_averager), but the hivemind machinery
+(DHT, matchmaking, butterfly all-reduce over libp2p, background averager
+processes) is replaced by a fixed single-node world: one process per GPU,
+one flat fp32 RCCL all-reduce over xGMI per outer round on a side HIP
+stream, fused HIP kernels for pseudo-gradient + Nesterov + copy-back.
+
+Mapping of reference concepts:
+  - a "peer"/"worker" = one torch.distributed rank (1 GPU per worker,
+    BASELINE.json configs[4]);
+  - DHT progress gossip (DiloCoProgressTracker, hivemind_diloco.py:174-282)
+    = a local step counter (all ranks step in lock-step, so global progress
+    IS local progress);
+  - WAIT_FOR_ALL/NO_WAIT straggler policy (:285-300) = inert enum kept for
+    CLI compatibility (RCCL collectives are synchronous on a fixed world);
+  - load_state_from_peers (train_fsdp.py:348-349) = broadcast of the flat
+    parameter/optimizer state from rank 0.
+"""
+
+from __future__ import annotations
+
+import contextlib
+from dataclasses import dataclass
+from enum import Enum
+from functools import partial
+from typing import Callable, Optional
+
+import torch
+import torch.distributed as dist
+
+from opendiloco_amd.optim import FlatSGDNesterov, FlatSpace, FusedAdamW
+
+
+class AllReduceStrategy(Enum):
+    """Kept for CLI compatibility (reference hivemind_diloco.py:285-297);
+    single-node RCCL is always effectively WAIT_FOR_ALL."""
+
+    WAIT_FOR_ALL = "WAIT_FOR_ALL"
+    NO_WAIT = "NO_WAIT"
+
+
+DEFAULT_TIMEOUT_WAITING_FOR_PEERS = 600
+
+
+def _world_size() -> int:
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+@dataclass
+class LocalTrainingProgress:
+    epoch: int = 0
+    samples_accumulated: int = 0
+
+
+@dataclass
+class GlobalTrainingProgress:
+    epoch: int = 0
+    num_peers: int = 1
+
+
+class DiloCoProgressTracker:
+    """Local step-counting replacement for the DHT tracker
+    (hivemind_diloco.py:174-282): an epoch ends when samples_accumulated
+    reaches batch_size * num_inner_steps (:184-191)."""
+
+    def __init__(self, batch_size: int, num_inner_steps: int):
+        self.batch_size = batch_size
+        self.num_inner_steps = num_inner_steps
+        self.target_batch_size = batch_size * num_inner_steps
+        self.local_progress = LocalTrainingProgress()
+        self.global_progress = GlobalTrainingProgress(num_peers=_world_size())
+
+    @property
+    def ready_to_update_epoch(self) -> bool:
+        return (self.global_progress.epoch > self.local_progress.epoch
+                or self.local_progress.samples_accumulated >= self.target_batch_size)
+
+    @property
+    def local_step(self) -> int:
+        return self.local_progress.samples_accumulated // self.batch_size
+
+    @property
+    def real_step(self) -> int:
+        return self.local_step + self.local_progress.epoch * self.num_inner_steps
+
+    def report_local_progress(self, epoch: int, samples_accumulated: int) -> None:
+        self.local_progress.epoch = epoch
+        self.local_progress.samples_accumulated = samples_accumulated
+
+    def update_epoch(self, new_epoch: int) -> None:
+        self.local_progress.epoch = new_epoch
+        self.global_progress.epoch = new_epoch
+        self.local_progress.samples_accumulated = 0
+
+    def pause_updates(self):
+        return contextlib.nullcontext()
+
+
+class DiLoCoGradAverager:
+    """Pseudo-gradient buffer + the cross-worker average.
+
+    Replaces hivemind's DecentralizedAverager subclass
+    (hivemind_diloco.py:61-171): ``compute_and_load_pseudo_grad_into_averager``
+    computes theta_outer - theta_local (:158-167) into ONE flat fp32 buffer,
+    and ``step`` runs a single all-reduce (ReduceOp.AVG over RCCL; the
+    reference's butterfly all-reduce / train_diloco_torch.py:345 twin).
+    """
+
+    def __init__(self, flat: FlatSpace, flat_outer: torch.Tensor,
+                 comm_dtype: torch.dtype = torch.float32):
+        self.flat = flat
+        self.flat_outer = flat_outer
+        self.comm_dtype = comm_dtype
+        self.pseudo_grad = torch.zeros_like(flat_outer)
+        self._comm_stream = (torch.cuda.Stream() if flat_outer.device.type == "cuda" else None)
+        self.last_allreduce_seconds: float = 0.0
+
+    def compute_and_load_pseudo_grad_into_averager(self) -> None:
+        if self.flat_outer.device.type == "cuda":
+            from opendiloco_amd.ops import _ext
+
+            _ext().pseudo_grad(self.pseudo_grad, self.flat_outer, self.flat.flat_param)
+        else:
+            torch.sub(self.flat_outer, self.flat.flat_param, out=self.pseudo_grad)
+
+    def step(self, wait: bool = True, timeout=None, control=None, **_kw) -> None:
+        self.compute_and_load_pseudo_grad_into_averager()
+        self.all_reduce_()
+
+    def all_reduce_(self) -> None:
+        ws = _world_size()
+        if ws <= 1:
+            return
+        buf = self.pseudo_grad
+        reduced_in_lower_precision = self.comm_dtype != torch.float32
+        if reduced_in_lower_precision:
+            send = buf.to(self.comm_dtype)
+        else:
+            send = buf
+        backend = dist.get_backend()
+        if self._comm_stream is not None:
+            # side HIP stream: the all-reduce payload is ready; ordering with
+            # the consuming outer kernel is restored via stream waits
+            self._comm_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._comm_stream):
+                dist.all_reduce(send, op=dist.ReduceOp.AVG)
+            torch.cuda.current_stream().wait_stream(self._comm_stream)
+        else:
+            if backend == "gloo":
+                dist.all_reduce(send, op=dist.ReduceOp.SUM)
+                send.div_(ws)
+            else:
+                dist.all_reduce(send, op=dist.ReduceOp.AVG)
+        if reduced_in_lower_precision:
+            buf.copy_(send)
+
+    def notify_used_averaged_gradients(self) -> None:
+        pass
+
+
+class DiLoCoStateAverager:
+    """Holds theta_outer (GPU-resident flat fp32 — the reference offloads to
+    CPU, hivemind_diloco.py:400; with 288 GB HBM3E per MI355X there is no
+    reason to) and the outer optimizer; carries the inner scheduler like the
+    reference's DiLoCoStateAverager (hivemind_diloco.py:35-58)."""
+
+    def __init__(self, flat: FlatSpace, outer_kwargs: dict, inner_optimizer,
+                 scheduler: Optional[Callable] = None):
+        self.flat = flat
+        self.main_parameters = flat.params
+        self.flat_outer = flat.flat_param.detach().clone()
+        self.optimizer = FlatSGDNesterov(self.flat_outer, **outer_kwargs)
+        self.local_epoch = 0
+        self.inner_optimizer = inner_optimizer
+        self.scheduler_inner_optimizer = scheduler(inner_optimizer) if scheduler is not None else None
+
+
+def _factory_kwargs(factory, expect_cls, what: str) -> dict:
+    """Extract hyperparameters from a functools.partial of a torch optimizer
+    (the boundary contract: honour hyperparams, not the torch impl —
+    reference passes partial(torch.optim.AdamW/SGD, ...), train_fsdp.py:250,253)."""
+    if isinstance(factory, partial):
+        return dict(factory.keywords)
+    raise TypeError(f"{what} must be a functools.partial of {expect_cls.__name__} "
+                    f"(got {type(factory)}); e.g. partial(torch.optim.AdamW, lr=4e-4)")
+
+
+class DiLoCoOptimizer:
+    """See module docstring.  Constructor signature mirrors
+    hivemind_diloco.py:326-343; DHT/compression args are accepted and inert
+    where single-node RCCL makes them meaningless."""
+
+    def __init__(self, *,
+                 dht=None,
+                 run_id: str = "diloco",
+                 batch_size: int,
+                 num_inner_steps: int,
+                 outer_optimizer,
+                 inner_optimizer,
+                 params=None,
+                 scheduler: Optional[Callable] = None,
+                 averager_opts: Optional[dict] = None,
+                 grad_compression=None,
+                 tracker_opts: Optional[dict] = None,
+                 all_reduce_strategy: AllReduceStrategy = AllReduceStrategy.WAIT_FOR_ALL,
+                 timeout_waiting_for_peers: Optional[float] = None,
+                 matchmaking_time: Optional[float] = 15.0,
+                 **kwargs):
+        # mirrored validation (hivemind_diloco.py:345-361,408-444)
+        if timeout_waiting_for_peers is not None and all_reduce_strategy == AllReduceStrategy.NO_WAIT:
+            raise ValueError("You cannot use timeout_waiting_for_peers with NO_WAIT strategy")
+        if (timeout_waiting_for_peers is not None and matchmaking_time is not None
+                and timeout_waiting_for_peers < matchmaking_time):
+            raise ValueError("timeout_waiting_for_peers must be greater than matchmaking_time")
+        for bad in ("optimizer", "target_batch_size", "batch_size_per_step"):
+            if bad in kwargs:
+                raise KeyError(f"{bad} should not be passed to DiLoCoOptimizer")
+        kwargs.pop("use_local_updates", None)
+        kwargs.pop("offload_optimizer", None)
+
+        self.run_id = run_id
+        self.batch_size = batch_size
+        self.num_inner_steps = num_inner_steps
+        self.all_reduce_strategy = all_reduce_strategy
+        self.timeout_waiting_for_peers = (timeout_waiting_for_peers
+                                          if timeout_waiting_for_peers is not None
+                                          else DEFAULT_TIMEOUT_WAITING_FOR_PEERS)
+        self.matchmaking_time = matchmaking_time
+
+        params = list(params)
+        if len(params) and isinstance(params[0], dict):
+            flat_params = [p for g in params for p in g["params"]]
+        else:
+            flat_params = params
+
+        # inner optimizer: fused flat AdamW with the factory's hyperparams
+        if isinstance(inner_optimizer, torch.optim.Optimizer):
+            raise TypeError("pass inner_optimizer as a factory (functools.partial), like the reference CLI")
+        inner_kwargs = _factory_kwargs(inner_optimizer, torch.optim.AdamW, "inner_optimizer")
+        self.inner_optimizer = FusedAdamW(flat_params, **inner_kwargs)
+        self.flat = self.inner_optimizer.flat
+
+        outer_kwargs = _factory_kwargs(outer_optimizer, torch.optim.SGD, "outer_optimizer")
+        self.state_averager = DiLoCoStateAverager(self.flat, outer_kwargs,
+                                                  self.inner_optimizer, scheduler)
+
+        comm_dtype = torch.float32
+        name = type(grad_compression).__name__ if grad_compression is not None else ""
+        if isinstance(grad_compression, str):
+            name = grad_compression
+        if "fp16" in name.lower() or "float16" in name.lower():
+            comm_dtype = torch.float16
+        self.diloco_grad_averager = DiLoCoGradAverager(self.flat, self.state_averager.flat_outer,
+                                                       comm_dtype=comm_dtype)
+
+        tracker_opts = dict(tracker_opts or {})
+        self.tracker = DiloCoProgressTracker(batch_size=batch_size, num_inner_steps=num_inner_steps)
+        self.scheduled_diloco_grads = None  # compat attr (hivemind_diloco.py:389)
+
+    # ---- properties mirroring the reference ----
+    @property
+    def local_epoch(self) -> int:
+        return self.state_averager.local_epoch
+
+    @property
+    def param_groups(self):
+        """Inner optimizer is the main optimizer (hivemind_diloco.py:692-695)."""
+        return self.inner_optimizer.param_groups
+
+    # ---- training step (hivemind_diloco.py:483-558) ----
+    def step(self, closure: Optional[Callable] = None, batch_size: Optional[int] = None,
+             scaler: Optional[torch.amp.GradScaler] = None):
+        if scaler is not None and closure is not None:
+            raise ValueError("You cannot use closure and scaler at the same time")
+        batch_size = batch_size if batch_size is not None else self.batch_size
+
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+
+        new_samples = self.tracker.local_progress.samples_accumulated + batch_size
+        self.tracker.report_local_progress(self.local_epoch, samples_accumulated=new_samples)
+
+        if scaler is not None:
+            scaler.step(self.inner_optimizer)
+        else:
+            self.inner_optimizer.step()
+
+        if self.state_averager.scheduler_inner_optimizer is not None:
+            self.state_averager.scheduler_inner_optimizer.step()
+
+        if self.tracker.ready_to_update_epoch:
+            self._update_global_epoch()
+        return loss
+
+    def _update_global_epoch(self) -> None:
+        """Outer round (hivemind_diloco.py:570-679): pseudo-grad, all-reduce,
+        Nesterov on theta_outer, copy-back to theta_local."""
+        import time
+
+        with self.tracker.pause_updates():
+            t0 = time.perf_counter()
+            self.diloco_grad_averager.step(wait=True)
+            self.state_averager.optimizer.step_fused(self.flat.flat_param,
+                                                     self.diloco_grad_averager.pseudo_grad)
+            self.state_averager.local_epoch += 1
+            self.tracker.update_epoch(self.state_averager.local_epoch)
+            self.diloco_grad_averager.last_allreduce_seconds = time.perf_counter() - t0
+
+    def zero_grad(self, set_to_none: bool = True) -> None:
+        self.inner_optimizer.zero_grad(set_to_none)
+
+    # ---- state (hivemind_diloco.py:697-714) ----
+    def state_dict(self) -> dict:
+        state_dict_outer = self.state_averager.optimizer.state_dict()
+        state_dict_outer["state"]["local_epoch"] = self.local_epoch
+        return {
+            "state_dict_outer": state_dict_outer,
+            "state_dict_inner": self.inner_optimizer.state_dict(),
+        }
+
+    def load_state_dict(self, state_dict: dict) -> None:
+        outer = state_dict["state_dict_outer"]
+        if "local_epoch" in outer["state"]:
+            self.state_averager.local_epoch = outer["state"].pop("local_epoch")
+        self.state_averager.optimizer.load_state_dict(outer)
+        self.inner_optimizer.load_state_dict(state_dict["state_dict_inner"])
+
+    def sync_outer_from_local(self) -> None:
+        """theta_outer := theta_local.  Mirrors the reference's resume
+        semantics where the offloaded outer copy is (re)made from the loaded
+        model parameters (train_fsdp.py:262-274 double-load dance)."""
+        self.state_averager.flat_outer.copy_(self.flat.flat_param)
+
+    def load_state_from_peers(self, **_kw) -> None:
+        """Broadcast flat parameter + optimizer state from rank 0 (replaces
+        hivemind state download, train_fsdp.py:348-349)."""
+        if _world_size() <= 1:
+            return
+        for t in (self.flat.flat_param, self.inner_optimizer.flat_m,
+                  self.inner_optimizer.flat_v, self.state_averager.flat_outer):
+            dist.broadcast(t, src=0)
+        if self.state_averager.optimizer.momentum_buf is not None:
+            dist.broadcast(self.state_averager.optimizer.momentum_buf, src=0)
+
+    def update_main_param_after_outer_step(self) -> None:
+        """theta_local := theta_outer (hivemind_diloco.py:716-720); already
+        fused into the outer kernel — kept for API compatibility."""
+        self.flat.flat_param.copy_(self.state_averager.flat_outer)

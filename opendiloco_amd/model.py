@@ -1,0 +1,229 @@
+"""MI355X-native Llama for causal LM.
+
+Same architecture, parameter names and initialisation as the
+``transformers.LlamaForCausalLM`` the reference trains
+(open_diloco/train_fsdp.py:171-174, train_diloco_torch.py:183): embedding ->
+N x (RMSNorm, QKV proj, RoPE, causal SDPA, O proj, RMSNorm, SwiGLU MLP) ->
+final RMSNorm -> lm_head -> shifted mean CE.  state_dict keys are identical
+to HF's so the 2m/150m/1b fixtures and checkpoints interchange.
+
+Compute policy (mirrors the reference's FSDP MixedPrecision(param_dtype=bf16)
++ fp32 master params, train_fsdp.py:239-245):
+  - parameters are fp32 masters;
+  - on GPU, forward casts each weight to ``compute_dtype`` (bf16 default)
+    inside autograd, runs hand-written HIP kernels for RMSNorm/RoPE/
+    attention/SwiGLU/CE and rocBLAS bf16 GEMMs (fp32 accumulate) for the
+    dense projections; gradients flow back to fp32 masters;
+  - on CPU (the reference's own CPU path / gloo tests), compute_dtype is
+    fp32 and the ops run plain torch fp32 math.
+"""
+
+from __future__ import annotations
+
+import math
+import os
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from opendiloco_amd import ops
+from opendiloco_amd.llama_config import LlamaModelConfig
+
+
+@dataclass
+class CausalLMOutput:
+    loss: torch.Tensor | None
+    logits: torch.Tensor | None
+
+
+def _cast(w: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    return w if w.dtype == dtype else w.to(dtype)
+
+
+class CastLinear(nn.Module):
+    """Bias-free linear with an fp32 master weight cast to compute dtype per
+    forward (autocast semantics); the GEMM itself is a rocBLAS bf16 GEMM with
+    fp32 accumulate — a plain library GEMM, per the MFMA design rules."""
+
+    def __init__(self, in_features: int, out_features: int):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.in_features, self.out_features = in_features, out_features
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F.linear(x, _cast(self.weight, x.dtype))
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden))
+        self.variance_epsilon = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.rmsnorm(x, _cast(self.weight, x.dtype), self.variance_epsilon)
+
+
+class Rotary:
+    """Precomputed fp32 cos/sin tables [S, D/2] (HF inv_freq convention)."""
+
+    def __init__(self, cfg: LlamaModelConfig):
+        D = cfg.head_dim
+        inv_freq = 1.0 / (cfg.rope_theta ** (torch.arange(0, D, 2, dtype=torch.float32) / D))
+        t = torch.arange(cfg.max_position_embeddings, dtype=torch.float32)
+        freqs = torch.outer(t, inv_freq)  # [S, D/2]
+        self.cos = freqs.cos()
+        self.sin = freqs.sin()
+
+    def to_(self, device):
+        self.cos = self.cos.to(device)
+        self.sin = self.sin.to(device)
+        return self
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaModelConfig):
+        super().__init__()
+        h, D = cfg.hidden_size, cfg.head_dim
+        self.num_heads = cfg.num_attention_heads
+        self.num_kv_heads = cfg.num_key_value_heads
+        self.head_dim = D
+        self.q_proj = CastLinear(h, self.num_heads * D)
+        self.k_proj = CastLinear(h, self.num_kv_heads * D)
+        self.v_proj = CastLinear(h, self.num_kv_heads * D)
+        self.o_proj = CastLinear(self.num_heads * D, h)
+        self.scale = D ** -0.5
+
+    def forward(self, x: torch.Tensor, rotary: Rotary) -> torch.Tensor:
+        B, S, _ = x.shape
+        D = self.head_dim
+        q = self.q_proj(x).view(B, S, self.num_heads, D).transpose(1, 2).contiguous()
+        k = self.k_proj(x).view(B, S, self.num_kv_heads, D).transpose(1, 2).contiguous()
+        v = self.v_proj(x).view(B, S, self.num_kv_heads, D).transpose(1, 2).contiguous()
+        q = ops.rope(q, rotary.cos, rotary.sin, S)
+        k = ops.rope(k, rotary.cos, rotary.sin, S)
+        o = ops.attention(q, k, v, self.scale)  # [B, Hq, S, D]
+        o = o.transpose(1, 2).reshape(B, S, self.num_heads * D)
+        return self.o_proj(o)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaModelConfig):
+        super().__init__()
+        h, i = cfg.hidden_size, cfg.intermediate_size
+        self.gate_proj = CastLinear(h, i)
+        self.up_proj = CastLinear(h, i)
+        self.down_proj = CastLinear(i, h)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaModelConfig):
+        super().__init__()
+        self.self_attn = Attention(cfg)
+        self.mlp = MLP(cfg)
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+
+    def forward(self, x: torch.Tensor, rotary: Rotary) -> torch.Tensor:
+        x = x + self.self_attn(self.input_layernorm(x), rotary)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaBackbone(nn.Module):
+    def __init__(self, cfg: LlamaModelConfig):
+        super().__init__()
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(DecoderLayer(cfg) for _ in range(cfg.num_hidden_layers))
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+
+
+class LlamaForCausalLM(nn.Module):
+    """Drop-in for the reference's model object: call signature
+    model(input_ids=..., attention_mask=..., labels=...) -> .loss/.logits."""
+
+    def __init__(self, config: LlamaModelConfig):
+        super().__init__()
+        self.config = config
+        self.model = LlamaBackbone(config)
+        self.lm_head = CastLinear(config.hidden_size, config.vocab_size)
+        if config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.rotary = Rotary(config)
+        self.compute_dtype: torch.dtype | None = None  # None -> fp32 on CPU, bf16 on GPU
+
+    # ---- init / load ----
+    def init_weights(self, seed: int | None = None):
+        """HF _init_weights: Linear/Embedding ~ N(0, initializer_range), norms = 1."""
+        if seed is not None:
+            torch.manual_seed(seed)
+        std = self.config.initializer_range
+        for m in self.modules():
+            if isinstance(m, CastLinear):
+                m.weight.data.normal_(0.0, std)
+            elif isinstance(m, nn.Embedding):
+                m.weight.data.normal_(0.0, std)
+            elif isinstance(m, RMSNorm):
+                m.weight.data.fill_(1.0)
+        return self
+
+    @classmethod
+    def from_pretrained(cls, path: str) -> "LlamaForCausalLM":
+        cfg = LlamaModelConfig.from_json(path)
+        model = cls(cfg)
+        # load HF safetensors if present (same state_dict keys)
+        import glob
+
+        files = sorted(glob.glob(os.path.join(path, "*.safetensors")))
+        if files:
+            from safetensors.torch import load_file
+
+            sd = {}
+            for f in files:
+                sd.update(load_file(f))
+            missing, unexpected = model.load_state_dict(sd, strict=False)
+            missing = [k for k in missing if "rotary" not in k]
+            if cfg.tie_word_embeddings:
+                missing = [k for k in missing if k != "lm_head.weight"]
+            assert not missing and not unexpected, (missing, unexpected)
+            model.float()
+        else:
+            model.init_weights()
+        return model
+
+    def _dtype_for(self, device: torch.device) -> torch.dtype:
+        if self.compute_dtype is not None:
+            return self.compute_dtype
+        return torch.float32 if device.type == "cpu" else torch.bfloat16
+
+    def forward(self, input_ids: torch.Tensor, attention_mask: torch.Tensor | None = None,
+                labels: torch.Tensor | None = None, **_ignored) -> CausalLMOutput:
+        device = input_ids.device
+        cdtype = self._dtype_for(device)
+        if self.rotary.cos.device != device:
+            self.rotary.to_(device)
+        S = input_ids.shape[1]
+        assert S <= self.config.max_position_embeddings
+
+        h = F.embedding(input_ids, _cast(self.model.embed_tokens.weight, cdtype))
+        for layer in self.model.layers:
+            h = layer(h, self.rotary)
+        h = self.model.norm(h)
+        logits = self.lm_head(h)  # [B, S, V], compute dtype
+
+        loss = None
+        if labels is not None:
+            # HF shift: predict token t+1 from logits at t; mean over B*(S-1)
+            V = logits.shape[-1]
+            shift_logits = logits[:, :-1, :].reshape(-1, V)
+            shift_labels = labels[:, 1:].reshape(-1)
+            loss = ops.causal_lm_loss(shift_logits.contiguous(), shift_labels.contiguous())
+        return CausalLMOutput(loss=loss, logits=logits)
+
+    def train_config_summary(self) -> dict:
+        return {"params": sum(p.numel() for p in self.parameters()), **self.config.to_dict()}

@@ -1,0 +1,268 @@
+"""Autograd wrappers over the gfx950 HIP kernels.
+
+Device policy (stated in DESIGN.md):
+  - On a GPU ("cuda" device on ROCm) every op REQUIRES the HIP extension and
+    fails loudly if it is missing — there is no silent eager fallback on the
+    product path.
+  - On CPU the same ops run as plain torch fp32 math.  This is the device the
+    reference's own CPU path uses (BASELINE.json configs[0]) and what the
+    multi-process gloo tests exercise; it is NOT used on a GPU box.
+
+Each op cites the reference math it replaces (transformers Llama internals
+invoked via model(**batch) at open_diloco/train_fsdp.py:378).
+"""
+
+from __future__ import annotations
+
+import torch
+
+_C = None
+_C_ERR: str | None = None
+
+
+def _ext():
+    """The HIP extension module; raises ImportError with instructions if absent."""
+    global _C, _C_ERR
+    if _C is None and _C_ERR is None:
+        try:
+            from opendiloco_amd.build_ext import load_binding
+
+            _C = load_binding()
+        except Exception as e:  # noqa: BLE001
+            _C_ERR = str(e)
+    if _C is None:
+        raise ImportError(
+            f"opendiloco_amd HIP extension unavailable on a GPU device — the GPU "
+            f"path never falls back to eager torch. Build with "
+            f"`python -m opendiloco_amd.build_ext`. Original error: {_C_ERR}"
+        )
+    return _C
+
+
+def ext_available() -> bool:
+    try:
+        _ext()
+        return True
+    except ImportError:
+        return False
+
+
+# ====================== RMSNorm ======================
+# transformers LlamaRMSNorm: fp32 variance, y = w * (x * rsqrt(mean(x^2)+eps))
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        if x.is_cuda:
+            y, invrms = _ext().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+            ctx.save_for_backward(x, w, invrms)
+            ctx.eps = eps
+            return y
+        xf = x.float()
+        var = xf.pow(2).mean(-1, keepdim=True)
+        invrms = torch.rsqrt(var + eps)
+        xn = (xf * invrms).to(x.dtype)
+        ctx.save_for_backward(x, w, invrms.squeeze(-1))
+        ctx.eps = eps
+        return w * xn
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, invrms = ctx.saved_tensors
+        if x.is_cuda:
+            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), x, w, invrms)
+            return dx, dw.to(w.dtype), None
+        xf, dyf, wf = x.float(), dy.float(), w.float()
+        ir = invrms.reshape(*x.shape[:-1], 1).float()
+        xhat = xf * ir
+        g = dyf * wf
+        dot = (g * xhat).mean(-1, keepdim=True)
+        dx = ((g - xhat * dot) * ir).to(x.dtype)
+        dw = (dyf * xhat).reshape(-1, x.shape[-1]).sum(0).to(w.dtype)
+        return dx, dw, None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    return _RMSNormFn.apply(x, w, eps)
+
+
+# ====================== RoPE ======================
+# transformers apply_rotary_pos_emb, half-split convention:
+#   rotate_half(x) = cat(-x[..., D/2:], x[..., :D/2])
+#   out = x*cos + rotate_half(x)*sin
+
+class _RopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, costab, sintab, S):
+        ctx.S = S
+        ctx.tabs = (costab, sintab)
+        if x.is_cuda:
+            return _ext().rope(x.contiguous(), costab, sintab, S, False)
+        return _rope_cpu(x, costab, sintab, S, backward=False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        costab, sintab = ctx.tabs
+        if dy.is_cuda:
+            dx = _ext().rope(dy.contiguous(), costab, sintab, ctx.S, True)
+        else:
+            dx = _rope_cpu(dy, costab, sintab, ctx.S, backward=True)
+        return dx, None, None, None
+
+
+def _rope_cpu(x, costab, sintab, S, backward):
+    D = x.shape[-1]
+    pos_shape = x.shape[:-1]
+    n = x.numel() // D
+    xf = x.float().reshape(n, D)
+    pos = (torch.arange(n) % S)
+    c = costab[pos]  # [n, D/2]
+    s = (-sintab[pos]) if backward else sintab[pos]
+    x1, x2 = xf[:, : D // 2], xf[:, D // 2:]
+    o1 = x1 * c - x2 * s
+    o2 = x2 * c + x1 * s
+    return torch.cat([o1, o2], dim=-1).reshape(*pos_shape, D).to(x.dtype)
+
+
+def rope(x: torch.Tensor, costab: torch.Tensor, sintab: torch.Tensor, S: int) -> torch.Tensor:
+    return _RopeFn.apply(x, costab, sintab, S)
+
+
+# ====================== SwiGLU ======================
+# transformers LlamaMLP: down( silu(gate(x)) * up(x) )
+
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        ctx.save_for_backward(gate, up)
+        if gate.is_cuda:
+            return _ext().swiglu_fwd(gate.contiguous(), up.contiguous())
+        return (torch.nn.functional.silu(gate.float()) * up.float()).to(gate.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        gate, up = ctx.saved_tensors
+        if gate.is_cuda:
+            dgate, dup = _ext().swiglu_bwd(dy.contiguous(), gate, up)
+            return dgate, dup
+        g, u, d = gate.float(), up.float(), dy.float()
+        sig = torch.sigmoid(g)
+        dgate = (d * u * (sig * (1 + g * (1 - sig)))).to(gate.dtype)
+        dup = (d * g * sig).to(up.dtype)
+        return dgate, dup
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    return _SwiGLUFn.apply(gate, up)
+
+
+# ====================== causal-LM cross entropy ======================
+# transformers LlamaForCausalLM loss: logits -> fp32, shift, CE mean.
+# Caller passes ALREADY-SHIFTED logits [T, V] and labels [T].
+
+class _CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels):
+        V = logits.shape[-1]
+        T = logits.numel() // V
+        if logits.is_cuda:
+            loss_rows, lse = _ext().ce_fwd(logits.contiguous(), labels.contiguous())
+            ctx.save_for_backward(logits, lse, labels)
+            ctx.T = T
+            return loss_rows.mean()
+        lf = logits.float()
+        lse = torch.logsumexp(lf, dim=-1)
+        loss_rows = lse - lf.gather(-1, labels.unsqueeze(-1)).squeeze(-1)
+        ctx.save_for_backward(logits, lse, labels)
+        ctx.T = T
+        return loss_rows.mean()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, lse, labels = ctx.saved_tensors
+        T = ctx.T
+        if logits.is_cuda:
+            dlogits = _ext().ce_bwd(logits, lse, labels, dloss.contiguous().float(), 1.0 / T)
+            return dlogits, None
+        p = torch.softmax(logits.float(), dim=-1)
+        p.scatter_add_(-1, labels.unsqueeze(-1), torch.full_like(labels, -1, dtype=p.dtype).unsqueeze(-1))
+        return (p * (dloss.float() / T)).to(logits.dtype), None
+
+
+def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Mean CE over pre-shifted rows (logits [T,V] vs labels [T])."""
+    return _CrossEntropyFn.apply(logits, labels)
+
+
+# ====================== attention ======================
+# Causal SDPA, layout [B, H, S, D]; GQA via Hkv < Hq.
+
+class _AttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        if q.is_cuda:
+            o, lse = _ext().attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+            ctx.save_for_backward(q, k, v, o, lse)
+            ctx.scale = scale
+            return o
+        o, lse = _attn_cpu_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        scale = ctx.scale
+        Hq, Hkv = q.shape[1], k.shape[1]
+        if q.is_cuda:
+            dq, dk_full, dv_full = _ext().attn_bwd(do.contiguous(), q, k, v, o, lse, scale)
+        else:
+            dq, dk_full, dv_full = _attn_cpu_bwd(do, q, k, v, o, lse, scale)
+        if Hq != Hkv:
+            g = Hq // Hkv
+            B, _, S, D = q.shape
+            dk = dk_full.view(B, Hkv, g, S, D).sum(2)
+            dv = dv_full.view(B, Hkv, g, S, D).sum(2)
+        else:
+            dk, dv = dk_full, dv_full
+        return dq, dk, dv, None
+
+
+def _attn_cpu_fwd(q, k, v, scale):
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    g = Hq // Hkv
+    kx = k.repeat_interleave(g, dim=1) if g > 1 else k
+    vx = v.repeat_interleave(g, dim=1) if g > 1 else v
+    s = (q.float() @ kx.float().transpose(-1, -2)) * scale
+    mask = torch.triu(torch.ones(S, S, dtype=torch.bool), diagonal=1)
+    s = s.masked_fill(mask, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.exp(s - lse.unsqueeze(-1))
+    o = (p @ vx.float()).to(q.dtype)
+    return o, lse
+
+
+def _attn_cpu_bwd(do, q, k, v, o, lse, scale):
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    g = Hq // Hkv
+    kx = k.repeat_interleave(g, dim=1) if g > 1 else k
+    vx = v.repeat_interleave(g, dim=1) if g > 1 else v
+    s = (q.float() @ kx.float().transpose(-1, -2)) * scale
+    mask = torch.triu(torch.ones(S, S, dtype=torch.bool), diagonal=1)
+    s = s.masked_fill(mask, float("-inf"))
+    p = torch.exp(s - lse.unsqueeze(-1).float())
+    dof = do.float()
+    dv_full = p.transpose(-1, -2) @ dof
+    dp = dof @ vx.float().transpose(-1, -2)
+    delta = (dof * o.float()).sum(-1, keepdim=True)
+    ds = p * (dp - delta) * scale
+    dq = (ds @ kx.float()).to(q.dtype)
+    dk_full = (ds.transpose(-1, -2) @ q.float()).to(q.dtype)
+    return dq, dk_full.to(q.dtype), dv_full.to(q.dtype)
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float) -> torch.Tensor:
+    return _AttentionFn.apply(q, k, v, scale)
