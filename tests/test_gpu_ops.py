@@ -1,0 +1,303 @@
+"""GPU kernel parity tests (MI355X): every HIP kernel against a plain
+PyTorch fp32 reference of the same op, on the same (bf16-rounded) inputs.
+
+Run: gpurun -- 'python -m pytest tests -m gpu -x -q'
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from opendiloco_amd.ops import _ext
+
+    return _ext()
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+def _to_dev_bf16(*ts):
+    return [t.to("cuda", torch.bfloat16).contiguous() for t in ts]
+
+
+# ---------------- MFMA layout probe ----------------
+
+@requires_gpu
+def test_mfma_probe_layout(ext):
+    """Verifies the A/B/C fragment maps assumed in attn.hip (asymmetric
+    operands per guide rule: symmetric inputs miss transposes)."""
+    a = torch.randn(16, 32)
+    b = torch.randn(32, 16) * torch.linspace(0.5, 2.0, 16)  # asymmetric
+    a16, b16 = _to_dev_bf16(a, b)
+    out = ext.probe_mfma(a16, b16)
+    ref = a16.float().cpu() @ b16.float().cpu()
+    assert torch.allclose(out.cpu(), ref, atol=1e-3, rtol=1e-3), \
+        f"max diff {(out.cpu()-ref).abs().max()}"
+
+
+# ---------------- RMSNorm ----------------
+
+@requires_gpu
+@pytest.mark.parametrize("rows,cols", [(64, 64), (512, 1024), (128, 2048), (33, 1024)])
+def test_rmsnorm_fwd_bwd(ext, rows, cols):
+    from opendiloco_amd import ops
+
+    x = torch.randn(rows, cols)
+    w = torch.randn(cols).abs() + 0.5
+    xg, wg = _to_dev_bf16(x, w)
+    xg.requires_grad_(True)
+    wg.requires_grad_(True)
+    y = ops.rmsnorm(xg, wg, 1e-5)
+    # fp32 reference on the same bf16-rounded values
+    xf, wf = xg.detach().float(), wg.detach().float()
+    ir = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5)
+    yref = wf * (xf * ir)
+    assert torch.allclose(y.float().cpu(), yref.cpu(), atol=3e-2, rtol=3e-2)
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    xf = xf.requires_grad_(True)
+    wf = wf.requires_grad_(True)
+    ir = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5)
+    (wf * (xf * ir)).backward(dy.float())
+    assert torch.allclose(xg.grad.float().cpu(), xf.grad.cpu(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(wg.grad.float().cpu(), wf.grad.cpu(), atol=5e-1, rtol=2e-2)
+
+
+# ---------------- RoPE ----------------
+
+@requires_gpu
+@pytest.mark.parametrize("D", [32, 64])
+def test_rope_fwd_bwd(ext, D):
+    from opendiloco_amd import ops
+
+    B, H, S = 2, 3, 128
+    inv_freq = 1.0 / (10000.0 ** (torch.arange(0, D, 2, dtype=torch.float32) / D))
+    freqs = torch.outer(torch.arange(S, dtype=torch.float32), inv_freq)
+    cos, sin = freqs.cos().cuda(), freqs.sin().cuda()
+    x = torch.randn(B, H, S, D)
+    (xg,) = _to_dev_bf16(x)
+    xg.requires_grad_(True)
+    y = ops.rope(xg, cos, sin, S)
+    xf = xg.detach().float()
+    c2 = torch.cat([cos, cos], -1).view(1, 1, S, D)
+    s2 = torch.cat([sin, sin], -1).view(1, 1, S, D)
+    rot = torch.cat([-xf[..., D // 2:], xf[..., : D // 2]], -1)
+    yref = xf * c2 + rot * s2
+    assert torch.allclose(y.float(), yref, atol=2e-2, rtol=2e-2)
+    # backward = transposed rotation; check vs autograd
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    xf.requires_grad_(True)
+    rot = torch.cat([-xf[..., D // 2:], xf[..., : D // 2]], -1)
+    (xf * c2 + rot * s2).backward(dy.float())
+    assert torch.allclose(xg.grad.float(), xf.grad, atol=2e-2, rtol=2e-2)
+
+
+# ---------------- SwiGLU ----------------
+
+@requires_gpu
+def test_swiglu_fwd_bwd(ext):
+    from opendiloco_amd import ops
+
+    gate, up = torch.randn(1024, 688), torch.randn(1024, 688)
+    gg, ug = _to_dev_bf16(gate, up)
+    gg.requires_grad_(True)
+    ug.requires_grad_(True)
+    y = ops.swiglu(gg, ug)
+    gf, uf = gg.detach().float().requires_grad_(True), ug.detach().float().requires_grad_(True)
+    yref = torch.nn.functional.silu(gf) * uf
+    assert torch.allclose(y.float(), yref, atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yref.backward(dy.float())
+    assert torch.allclose(gg.grad.float(), gf.grad, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(ug.grad.float(), uf.grad, atol=2e-2, rtol=2e-2)
+
+
+# ---------------- cross entropy ----------------
+
+@requires_gpu
+@pytest.mark.parametrize("T,V", [(256, 1024), (100, 32000), (7, 1000)])
+def test_cross_entropy_fwd_bwd(ext, T, V):
+    from opendiloco_amd import ops
+
+    logits = torch.randn(T, V) * 3
+    labels = torch.randint(0, V, (T,))
+    lg = logits.to("cuda", torch.bfloat16).contiguous().requires_grad_(True)
+    lab = labels.cuda()
+    loss = ops.causal_lm_loss(lg, lab)
+    lf = lg.detach().float().requires_grad_(True)
+    lref = torch.nn.functional.cross_entropy(lf, lab)
+    assert loss.item() == pytest.approx(lref.item(), rel=1e-3, abs=1e-3)
+    loss.backward()
+    lref.backward()
+    assert torch.allclose(lg.grad.float(), lf.grad, atol=1e-4, rtol=1e-2)
+
+
+# ---------------- attention ----------------
+
+@requires_gpu
+@pytest.mark.parametrize("B,Hq,Hkv,S,D", [
+    (2, 2, 2, 128, 32),     # 2m shape
+    (2, 4, 4, 256, 64),     # 150m shape (reduced)
+    (1, 16, 16, 1024, 64),  # 150m full seq
+    (2, 8, 2, 128, 64),     # GQA (1b shape)
+    (1, 2, 2, 96, 64),      # ragged tail (S % 64 != 0)
+])
+def test_attention_fwd_bwd(ext, B, Hq, Hkv, S, D):
+    from opendiloco_amd import ops
+
+    q = torch.randn(B, Hq, S, D)
+    k = torch.randn(B, Hkv, S, D)
+    v = torch.randn(B, Hkv, S, D)
+    qg, kg, vg = _to_dev_bf16(q, k, v)
+    for t in (qg, kg, vg):
+        t.requires_grad_(True)
+    scale = D ** -0.5
+    o = ops.attention(qg, kg, vg, scale)
+
+    # fp32 SDPA reference on the same bf16-rounded inputs
+    g = Hq // Hkv
+    kf = kg.detach().float().repeat_interleave(g, 1) if g > 1 else kg.detach().float()
+    vf = vg.detach().float().repeat_interleave(g, 1) if g > 1 else vg.detach().float()
+    qf = qg.detach().float().requires_grad_(True)
+    kf = kf.requires_grad_(True)
+    vf = vf.requires_grad_(True)
+    s = (qf @ kf.transpose(-1, -2)) * scale
+    mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device="cuda"), 1)
+    s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, -1)
+    oref = p @ vf
+    assert torch.allclose(o.float(), oref, atol=3e-2, rtol=3e-2), \
+        f"fwd max diff {(o.float()-oref).abs().max()}"
+
+    do = torch.randn_like(o)
+    o.backward(do)
+    oref.backward(do.float())
+    dkr = kf.grad
+    dvr = vf.grad
+    if g > 1:
+        dkr = dkr.view(B, Hkv, g, S, D).sum(2)
+        dvr = dvr.view(B, Hkv, g, S, D).sum(2)
+    for got, ref, name, tol in [(qg.grad, qf.grad, "dq", 5e-2), (kg.grad, dkr, "dk", 5e-2),
+                                (vg.grad, dvr, "dv", 5e-2)]:
+        md = (got.float() - ref).abs().max().item()
+        scale_ref = ref.abs().max().item() + 1e-6
+        assert md / scale_ref < tol, f"{name} rel max diff {md/scale_ref}"
+
+
+@requires_gpu
+def test_attention_deterministic(ext):
+    from opendiloco_amd import ops
+
+    q, k, v = _to_dev_bf16(torch.randn(2, 4, 256, 64), torch.randn(2, 4, 256, 64),
+                           torch.randn(2, 4, 256, 64))
+    for t in (q, k, v):
+        t.requires_grad_(True)
+    do = torch.randn(2, 4, 256, 64, device="cuda", dtype=torch.bfloat16)
+
+    def run():
+        for t in (q, k, v):
+            t.grad = None
+        o = ops.attention(q, k, v, 0.125)
+        o.backward(do)
+        return o.detach().clone(), q.grad.clone(), k.grad.clone(), v.grad.clone()
+
+    r1, r2 = run(), run()
+    for a, b in zip(r1, r2):
+        assert torch.equal(a, b)
+
+
+@requires_gpu
+def test_attention_f16(ext):
+    from opendiloco_amd import ops
+
+    q = torch.randn(1, 2, 128, 64, device="cuda", dtype=torch.float16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    o = ops.attention(q, k, v, 0.125)
+    s = (q.float() @ k.float().transpose(-1, -2)) * 0.125
+    s = s.masked_fill(torch.triu(torch.ones(128, 128, dtype=torch.bool, device="cuda"), 1),
+                      float("-inf"))
+    oref = torch.softmax(s, -1) @ v.float()
+    assert torch.allclose(o.float(), oref, atol=3e-2, rtol=3e-2)
+
+
+# ---------------- fused AdamW ----------------
+
+@requires_gpu
+def test_fused_adamw_matches_torch(ext):
+    n = 1_000_003  # odd size exercises the tail kernel
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    pref = torch.nn.Parameter(p.clone())
+    opt = torch.optim.AdamW([pref], lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.1,
+                            foreach=False)
+    for step in range(1, 4):
+        gg = torch.randn(n, device="cuda") if step > 1 else g
+        pref.grad = gg.clone()
+        opt.step()
+        ext.fused_adamw(p, gg, m, v, 1e-2, 0.9, 0.95, 1e-8, 0.1, step)
+        md = (p - pref.data).abs().max().item()
+        assert md < 1e-6, f"step {step}: {md}"
+
+
+# ---------------- clip / pseudo-grad / nesterov ----------------
+
+@requires_gpu
+def test_clip_matches_torch(ext):
+    n = 2_000_001
+    g = torch.randn(n, device="cuda") * 2
+    gref = g.clone()
+    out2 = ext.clip_grad_(g, 1.0)
+    total = gref.norm(2)
+    coef = (1.0 / (total + 1e-6)).clamp(max=1.0)
+    assert out2[0].item() == pytest.approx(total.item(), rel=1e-5)
+    assert torch.allclose(g, gref * coef, atol=1e-6)
+    # no-clip case: norm below max_norm leaves grads untouched
+    g2 = torch.randn(1000, device="cuda") * 1e-3
+    g2ref = g2.clone()
+    ext.clip_grad_(g2, 1.0)
+    assert torch.equal(g2, g2ref)
+
+
+@requires_gpu
+def test_pseudo_grad_and_nesterov(ext):
+    n = 500_001
+    outer = torch.randn(n, device="cuda")
+    local = torch.randn(n, device="cuda")
+    gbuf = torch.empty(n, device="cuda")
+    ext.pseudo_grad(gbuf, outer, local)
+    assert torch.equal(gbuf, outer - local)
+
+    # torch SGD nesterov reference over 3 steps
+    pref = torch.nn.Parameter(outer.clone())
+    sgd = torch.optim.SGD([pref], lr=0.7, momentum=0.9, nesterov=True, foreach=False)
+    buf = torch.empty(n, device="cuda")
+    o, l = outer.clone(), local.clone()
+    for step in range(3):
+        g = torch.randn(n, device="cuda")
+        pref.grad = g.clone()
+        sgd.step()
+        ext.outer_nesterov(o, l, buf, g, 0.7, 0.9, step == 0)
+        assert torch.allclose(o, pref.data, atol=1e-5), step
+        assert torch.equal(l, o)
+
+
+@requires_gpu
+def test_cast(ext):
+    src = torch.randn(12345, device="cuda")
+    dst = torch.empty(12345, device="cuda", dtype=torch.bfloat16)
+    ext.cast_(dst, src)
+    assert torch.allclose(dst.float(), src, atol=0.02, rtol=0.01)

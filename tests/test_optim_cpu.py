@@ -1,0 +1,117 @@
+"""CPU tests of the flat optimizers against torch's own implementations
+(the reference's inner/outer optimizers, train_diloco_torch.py:186-187)."""
+
+import copy
+
+import pytest
+import torch
+
+from opendiloco_amd.optim import FlatSGDNesterov, FlatSpace, FusedAdamW, clip_grad_norm_flat_
+
+
+def _toy_params(seed=0):
+    torch.manual_seed(seed)
+    return [torch.nn.Parameter(torch.randn(13, 7)), torch.nn.Parameter(torch.randn(29)),
+            torch.nn.Parameter(torch.randn(5, 5, 5))]
+
+
+def test_fused_adamw_matches_torch_adamw():
+    p1 = _toy_params()
+    p2 = [torch.nn.Parameter(p.data.clone()) for p in p1]
+    opt1 = FusedAdamW(p1, lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.1)
+    opt2 = torch.optim.AdamW(p2, lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.1,
+                             foreach=False)
+    for step in range(5):
+        torch.manual_seed(100 + step)
+        grads = [torch.randn_like(p) for p in p1]
+        for p, g in zip(p1, grads):
+            p.grad.copy_(g)
+        for p, g in zip(p2, grads):
+            p.grad = g.clone()
+        opt1.step()
+        opt2.step()
+        for a, b in zip(p1, p2):
+            assert torch.allclose(a.data, b.data, atol=1e-7), step
+    # state dict structure: per-param step/exp_avg/exp_avg_sq like torch
+    sd = opt1.state_dict()
+    assert set(sd["state"][0].keys()) == {"step", "exp_avg", "exp_avg_sq"}
+
+
+def test_fused_adamw_state_roundtrip():
+    p1 = _toy_params(1)
+    opt = FusedAdamW(p1, lr=1e-3, betas=(0.9, 0.95), weight_decay=0.1)
+    for _ in range(3):
+        for p in p1:
+            p.grad.copy_(torch.randn_like(p))
+        opt.step()
+    sd = copy.deepcopy(opt.state_dict())
+    p2 = [torch.nn.Parameter(p.data.clone()) for p in p1]
+    opt2 = FusedAdamW(p2, lr=1e-3, betas=(0.9, 0.95), weight_decay=0.1)
+    opt2.load_state_dict(sd)
+    assert torch.equal(opt.flat_m, opt2.flat_m)
+    assert torch.equal(opt.flat_v, opt2.flat_v)
+    assert opt2._step_count_t == 3
+    # continues identically
+    g = [torch.randn_like(p) for p in p1]
+    for p, gr in zip(p1, g):
+        p.grad.copy_(gr)
+    for p, gr in zip(p2, g):
+        p.grad.copy_(gr)
+    opt.step()
+    opt2.step()
+    for a, b in zip(p1, p2):
+        assert torch.equal(a.data, b.data)
+
+
+def test_flat_sgd_nesterov_matches_torch_sgd():
+    torch.manual_seed(2)
+    n = 1000
+    theta = torch.randn(n)
+    ref_p = torch.nn.Parameter(theta.clone())
+    ref = torch.optim.SGD([ref_p], lr=0.7, momentum=0.9, nesterov=True)
+    mine_outer = theta.clone()
+    mine_local = torch.randn(n)
+    sgd = FlatSGDNesterov(mine_outer, lr=0.7, momentum=0.9, nesterov=True)
+    for step in range(4):
+        g = torch.randn(n)
+        ref_p.grad = g.clone()
+        ref.step()
+        sgd.step_fused(mine_local, g)
+        assert torch.allclose(mine_outer, ref_p.data, atol=1e-6), step
+        assert torch.equal(mine_local, mine_outer)
+    # state dict carries the momentum buffer
+    sd = sgd.state_dict()
+    assert torch.allclose(sd["state"][0]["momentum_buffer"],
+                          ref.state[ref_p]["momentum_buffer"], atol=1e-6)
+    sgd2 = FlatSGDNesterov(mine_outer.clone(), lr=0.7)
+    sgd2.load_state_dict(sd)
+    assert torch.allclose(sgd2.momentum_buf, sgd.momentum_buf)
+
+
+def test_flatspace_views_and_zero_grad():
+    params = _toy_params(3)
+    datas = [p.data.clone() for p in params]
+    flat = FlatSpace(params)
+    for p, d in zip(params, datas):
+        assert torch.equal(p.data, d)
+        assert p.grad is not None and p.grad.shape == p.shape
+    params[0].grad.fill_(2.0)
+    assert flat.flat_grad[: params[0].numel()].eq(2).all()
+    flat.zero_grad()
+    assert flat.flat_grad.eq(0).all()
+
+
+def test_clip_flat_matches_torch_clip_cpu():
+    params = _toy_params(4)
+    flat = FlatSpace(params)
+    torch.manual_seed(9)
+    flat.flat_grad.copy_(torch.randn(flat.n) * 3)
+    ref_grads = [g.clone() for g in flat.grad_views()]
+    ref_params = [torch.nn.Parameter(torch.zeros_like(p)) for p in params]
+    for rp, rg in zip(ref_params, ref_grads):
+        rp.grad = rg
+    tn_ref = torch.nn.utils.clip_grad_norm_(ref_params, 1.0)
+    tn = clip_grad_norm_flat_(flat.flat_grad, 1.0)
+    assert tn.item() == pytest.approx(tn_ref.item(), rel=1e-6)
+    for g, rg in zip(flat.grad_views(), ref_grads):
+        assert torch.allclose(g, rg, atol=1e-6)
