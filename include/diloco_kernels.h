@@ -140,6 +140,8 @@ int dk_cast(void* dst, const void* src, int64_t n, int dst_dtype, int src_dtype,
  * kind: 0 = A-layout of mfma_f32_16x16x32_bf16, 1 = B-layout, 2 = C-layout. */
 int dk_probe_mfma_16x16x32_bf16(float* out_d, const void* a16x32, const void* b32x16,
                                 dkStream stream);
+int dk_probe_mfma_16x16x32_bf16_alt(float* out_d, const void* a16x32, const void* b32x16,
+                                    dkStream stream);
 
 /* version / build info */
 const char* dk_version(void);

@@ -685,3 +685,31 @@ extern "C" int dk_probe_mfma_16x16x32_bf16(float* out_d, const void* a16x32,
   DK_CHECK_LAUNCH();
   return 0;
 }
+
+// Alternative candidate mapping (k interleaved: k = (l>>4) + 4*j) — a
+// diagnostic twin so one GPU round can identify the true layout if the
+// primary assumption fails.
+__global__ void probe_mfma_alt_kernel(float* __restrict__ out,
+                                      const unsigned short* __restrict__ a,
+                                      const unsigned short* __restrict__ b) {
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 15, hi = lane >> 4;
+  shortx8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ((unsigned short*)&af)[j] = a[lo * 32 + (hi + 4 * j)];
+    ((unsigned short*)&bf)[j] = b[(hi + 4 * j) * 16 + lo];
+  }
+  floatx4 c = (floatx4)(0.f);
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) out[(hi * 4 + r) * 16 + lo] = c[r];
+}
+
+extern "C" int dk_probe_mfma_16x16x32_bf16_alt(float* out_d, const void* a16x32,
+                                               const void* b32x16, dkStream stream) {
+  hipLaunchKernelGGL(probe_mfma_alt_kernel, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                     out_d, (const unsigned short*)a16x32, (const unsigned short*)b32x16);
+  DK_CHECK_LAUNCH();
+  return 0;
+}

@@ -218,6 +218,13 @@ at::Tensor probe_mfma(const at::Tensor& a, const at::Tensor& b) {
   return out;
 }
 
+at::Tensor probe_mfma_alt(const at::Tensor& a, const at::Tensor& b) {
+  auto out = at::zeros({16, 16}, a.options().dtype(at::kFloat));
+  DK_OK(dk_probe_mfma_16x16x32_bf16_alt(out.data_ptr<float>(), a.data_ptr(), b.data_ptr(),
+                                        stream()));
+  return out;
+}
+
 std::string version() { return dk_version(); }
 
 }  // namespace
@@ -239,5 +246,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("outer_nesterov", &outer_nesterov);
   m.def("cast_", &cast_);
   m.def("probe_mfma", &probe_mfma);
+  m.def("probe_mfma_alt", &probe_mfma_alt);
   m.def("version", &version);
 }
