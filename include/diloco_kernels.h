@@ -40,9 +40,11 @@ int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy, const void* x,
                    const void* w, const float* invrms,
                    int64_t rows, int64_t cols, int grid, int dtype, dkStream stream);
 int dk_rmsnorm_bwd_grid(int64_t rows);
-/* out[c] = sum_g partial[g][c], fixed order (deterministic). out dtype f32. */
-int dk_reduce_partials(float* out, const float* partial, int grid, int64_t cols,
-                       dkStream stream);
+/* out[c] = sum_g partial[g][c], fixed order (deterministic), two parallel
+ * stages through tmp (f32[dk_reduce_tmp_slices(grid)][cols]). */
+int dk_reduce_tmp_slices(int grid);
+int dk_reduce_partials(float* out, float* tmp, const float* partial, int grid,
+                       int64_t cols, dkStream stream);
 
 /* ---- RoPE ---------------------------------------------------------------
  * Replaces transformers' apply_rotary_pos_emb (half-split convention:

@@ -56,12 +56,14 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
   const int64_t rows = x.numel() / cols;
   const int grid = dk_rmsnorm_bwd_grid(rows);
   auto dx = at::empty_like(x);
-  auto dwp = at::zeros({grid, cols}, x.options().dtype(at::kFloat));
+  auto dwp = at::empty({grid, cols}, x.options().dtype(at::kFloat));
+  auto tmp = at::empty({dk_reduce_tmp_slices(grid), cols}, x.options().dtype(at::kFloat));
   auto dw = at::empty({cols}, x.options().dtype(at::kFloat));
   DK_OK(dk_rmsnorm_bwd(dx.data_ptr(), dwp.data_ptr<float>(), dy.data_ptr(), x.data_ptr(),
                        w.data_ptr(), invrms.data_ptr<float>(), rows, cols, grid,
                        dt_of(x), stream()));
-  DK_OK(dk_reduce_partials(dw.data_ptr<float>(), dwp.data_ptr<float>(), grid, cols, stream()));
+  DK_OK(dk_reduce_partials(dw.data_ptr<float>(), tmp.data_ptr<float>(),
+                           dwp.data_ptr<float>(), grid, cols, stream()));
   return {dx, dw};
 }
 

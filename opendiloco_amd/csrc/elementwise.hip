@@ -70,8 +70,11 @@ __global__ void rmsnorm_fwd_kernel(typename DTraits<DT>::T* __restrict__ y,
 
 // backward: let xhat = x*invrms, g = dy*w.
 //   dx = (g - xhat * mean(g*xhat)) * invrms
-//   dw_partial[bid][c] += dy * xhat   (deterministic: block b owns rows b, b+grid, ...)
-template <int DT, int NT>
+//   dw partials accumulate in REGISTERS per block (thread t owns columns
+//   t, t+NT, ...) and are written once at the end — no global RMW, no
+//   zero-init, deterministic (block b owns rows b, b+grid, ...).
+// MAXC = max columns per thread: cols <= NT*MAXC (8192 at NT=256).
+template <int DT, int NT, int MAXC>
 __global__ void rmsnorm_bwd_kernel(typename DTraits<DT>::T* __restrict__ dx,
                                    float* __restrict__ dwp,
                                    const typename DTraits<DT>::T* __restrict__ dy,
@@ -82,7 +85,9 @@ __global__ void rmsnorm_bwd_kernel(typename DTraits<DT>::T* __restrict__ dx,
   using TR = DTraits<DT>;
   using T = typename TR::T;
   __shared__ float sred[NT / DK_WAVE];
-  float* dwrow = dwp + (int64_t)blockIdx.x * cols;
+  float dwacc[MAXC];
+#pragma unroll
+  for (int k = 0; k < MAXC; ++k) dwacc[k] = 0.f;
 
   for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
     const T* xr = x + r * cols;
@@ -96,24 +101,46 @@ __global__ void rmsnorm_bwd_kernel(typename DTraits<DT>::T* __restrict__ dx,
       dot += g * xh;
     }
     dot = block_reduce_sum<NT>(dot, sred) / (float)cols;
-    for (int c = threadIdx.x; c < cols; c += NT) {
+    int k = 0;
+    for (int c = threadIdx.x; c < cols; c += NT, ++k) {
       float xh = TR::toF(xr[c]) * ir;
       float dyf = TR::toF(dyr[c]);
       float g = dyf * TR::toF(w[c]);
       dxr[c] = TR::fromF((g - xh * dot) * ir);
-      dwrow[c] += dyf * xh;
+      dwacc[k] += dyf * xh;
     }
-    __syncthreads();  // dwrow reused next row iteration by same block only; sred reuse
+  }
+  {
+    int k = 0;
+    float* dwrow = dwp + (int64_t)blockIdx.x * cols;
+    for (int c = threadIdx.x; c < cols; c += NT, ++k) dwrow[c] = dwacc[k];
   }
 }
 
-__global__ void reduce_partials_kernel(float* __restrict__ out,
+// two-stage deterministic reduce of [grid][cols] -> [cols]:
+//  stage 1: block (y, cc) sums a 32-row slice -> tmp[y][cols]  (coalesced)
+//  stage 2: sums the <=64 tmp rows -> out[cols]
+__global__ void reduce_partials_stage1(float* __restrict__ tmp,
                                        const float* __restrict__ partial,
-                                       int grid, int64_t cols) {
+                                       int grid, int nslices, int64_t cols) {
+  const int y = blockIdx.y;
+  const int g0 = y * 32;
+  const int g1 = min(grid, g0 + 32);
   for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
        c += (int64_t)gridDim.x * blockDim.x) {
     float s = 0.f;
-    for (int g = 0; g < grid; ++g) s += partial[(int64_t)g * cols + c];
+    for (int g = g0; g < g1; ++g) s += partial[(int64_t)g * cols + c];
+    tmp[(int64_t)y * cols + c] = s;
+  }
+}
+
+__global__ void reduce_partials_stage2(float* __restrict__ out,
+                                       const float* __restrict__ tmp,
+                                       int nslices, int64_t cols) {
+  for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
+       c += (int64_t)gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int g = 0; g < nslices; ++g) s += tmp[(int64_t)g * cols + c];
     out[c] = s;
   }
 }
@@ -139,21 +166,34 @@ extern "C" int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy, const
                               const void* w, const float* invrms, int64_t rows,
                               int64_t cols, int grid, int dtype, dkStream stream) {
   constexpr int NT = 256;
+  if (cols > NT * 32) return (int)hipErrorInvalidValue;
   DK_DISPATCH_DT(dtype, {
     using T = typename DTraits<kDT>::T;
-    hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT>), dim3(grid), dim3(NT), 0,
-                       (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
-                       (const T*)x, (const T*)w, invrms, rows, (int)cols);
+    if (cols <= NT * 8)
+      hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 8>), dim3(grid), dim3(NT), 0,
+                         (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                         (const T*)x, (const T*)w, invrms, rows, (int)cols);
+    else
+      hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 32>), dim3(grid), dim3(NT), 0,
+                         (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                         (const T*)x, (const T*)w, invrms, rows, (int)cols);
   });
   DK_CHECK_LAUNCH();
   return 0;
 }
 
-extern "C" int dk_reduce_partials(float* out, const float* partial, int grid,
+// tmp must hold dk_reduce_tmp_slices(grid) * cols floats
+extern "C" int dk_reduce_tmp_slices(int grid) { return (grid + 31) / 32; }
+
+extern "C" int dk_reduce_partials(float* out, float* tmp, const float* partial, int grid,
                                   int64_t cols, dkStream stream) {
-  int blocks = dk_stream_grid(cols, 256);
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(blocks), dim3(256), 0,
-                     (hipStream_t)stream, out, partial, grid, cols);
+  const int nslices = dk_reduce_tmp_slices(grid);
+  int cblocks = dk_stream_grid(cols, 256);
+  hipLaunchKernelGGL(reduce_partials_stage1, dim3(cblocks, nslices), dim3(256), 0,
+                     (hipStream_t)stream, tmp, partial, grid, nslices, cols);
+  DK_CHECK_LAUNCH();
+  hipLaunchKernelGGL(reduce_partials_stage2, dim3(cblocks), dim3(256), 0,
+                     (hipStream_t)stream, out, tmp, nslices, cols);
   DK_CHECK_LAUNCH();
   return 0;
 }
