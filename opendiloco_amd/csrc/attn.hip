@@ -62,7 +62,27 @@ __device__ __forceinline__ float grp16_sum(float x) {
 }
 
 // ======================= forward =======================
-// LDS layout (dynamic): K[32][D+8] | V_T[D][32+8] | P[4][16][32+8]
+// v2 structure (guide T14/T3/T5): KT=64 kv tiles, double-buffered LDS, async
+// register staging (next tile's global loads issued before compute, written
+// to the other LDS buffer after it — HBM latency hides under MFMA), ONE
+// barrier per tile, XOR-swizzled transposed V image (kills the staging
+// write bank conflicts), s_setprio(1) around the MFMA clusters.
+// LDS: K[2][KT][DS] | V_T[2][D][KS (swizzled)] | P[4][16][KS]
+
+// byte-level XOR swizzle for the transposed-V image: row d's 16-B groups are
+// rotated by (d>>3)&7 — spreads the 8-u16-scatter staging writes over 8
+// banks and keeps every 16-B read aligned (XOR bits 4-6 only; row data span
+// 128 B < 144 B stride so the XOR never leaves the row).
+// xmask limits the XOR to 16-B groups that stay inside the row's data span:
+// 7 for a 128-B span (KS=72 u16 rows), 3 for a 64-B span (QS=40 u16 rows).
+template <typename T>
+__device__ __forceinline__ T* vt_addr(T* base, int d, int kbyte_off, int KSbytes,
+                                      int xmask = 7) {
+  int byte = d * KSbytes + kbyte_off;
+  byte ^= ((d >> 3) & xmask) << 4;
+  return (T*)((char*)base + byte);
+}
+
 template <int DT, int D>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     typename DTraits<DT>::T* __restrict__ o, float* __restrict__ lse,
@@ -74,16 +94,18 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   using T = typename TR::T;
   using MF = MFMA16<DT>;
   using frag = typename MF::frag;
-  constexpr int KT = 32;           // kv tile
-  constexpr int KS = KT + 8;       // padded LDS stride (keys dim)
+  constexpr int KT = 64;           // kv tile
+  constexpr int KS = KT + 8;       // padded LDS stride (keys dim), u16 units
   constexpr int DS = D + 8;        // padded LDS stride (channel dim)
   constexpr int NKC = D / 32;      // mfma K-chunks over channels
   constexpr int NDN = D / 16;      // output channel tiles
+  constexpr int NNT = KT / 16;     // 16-key sub-tiles per kv tile
+  constexpr int LPT = (KT * D) / 8 / 256;  // b128 loads per thread per tile
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  T* K_lds = (T*)smem_raw;                       // [KT][DS]
-  T* VT_lds = K_lds + KT * DS;                   // [D][KS]
-  T* P_lds = VT_lds + D * KS;                    // [4][16][KS]
+  T* K_lds = (T*)smem_raw;                       // [2][KT][DS]
+  T* VT_lds = K_lds + 2 * KT * DS;               // [2][D][KS] swizzled
+  T* P_lds = VT_lds + 2 * D * KS;                // [4][16][KS]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -101,8 +123,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
 
-  // Q A-fragments for this wave's 16 rows (row may exceed S on the tail
-  // tile: clamp the load address, rows >= S are masked out of all writes).
+  // Q A-fragments for this wave's 16 rows (clamped on the tail tile)
   frag q_frag[NKC];
   {
     const int qrow = q0 + lo;
@@ -122,77 +143,110 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int kv_end = min(S, qt * 64 + 64);       // causal upper bound
   const int n_kt = (kv_end + KT - 1) / KT;
 
+  // staging registers (async split): this thread's pieces of the next tile
+  frag kreg[LPT], vreg[LPT];
+  const int st_row[2] = {(int)threadIdx.x / (D / 8), (int)(threadIdx.x + 256) / (D / 8)};
+  const int st_c8[2] = {((int)threadIdx.x % (D / 8)) * 8, ((int)(threadIdx.x + 256) % (D / 8)) * 8};
+
+  auto load_tile = [&](int kt) {
+#pragma unroll
+    for (int i = 0; i < LPT; ++i) {
+      const int krow = kt * KT + st_row[i];
+      const int kr_c = krow < S ? krow : S - 1;
+      kreg[i] = *(const frag*)(k + kvoff + (int64_t)kr_c * D + st_c8[i]);
+      vreg[i] = *(const frag*)(v + kvoff + (int64_t)kr_c * D + st_c8[i]);
+    }
+  };
+  auto write_tile = [&](int buf) {
+    T* Kb = K_lds + buf * KT * DS;
+    T* Vb = VT_lds + buf * D * KS;
+#pragma unroll
+    for (int i = 0; i < LPT; ++i) {
+      *(frag*)(Kb + st_row[i] * DS + st_c8[i]) = kreg[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *vt_addr(Vb, st_c8[i] + j, st_row[i] * 2, KS * 2) = ((const T*)&vreg[i])[j];
+    }
+  };
+
+  load_tile(0);
+  write_tile(0);
+  __syncthreads();
+
   for (int kt = 0; kt < n_kt; ++kt) {
     const int kbase = kt * KT;
-    __syncthreads();
-    // ---- stage K tile (row-major, padded) + V tile (transposed) ----
-    {
-      constexpr int LPT = (KT * D) / 8;          // 16-B loads for the tile
-      for (int t = threadIdx.x; t < LPT; t += 256) {
-        const int row = t / (D / 8);
-        const int c8 = (t % (D / 8)) * 8;
-        const int krow = kbase + row;
-        const int kr_c = krow < S ? krow : S - 1;
-        frag kv8 = *(const frag*)(k + kvoff + (int64_t)kr_c * D + c8);
-        *(frag*)(K_lds + row * DS + c8) = kv8;
-        frag vv8 = *(const frag*)(v + kvoff + (int64_t)kr_c * D + c8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) VT_lds[(c8 + j) * KS + row] = ((const T*)&vv8)[j];
-      }
-    }
-    __syncthreads();
+    const int cur = kt & 1;
+    T* Kb = K_lds + cur * KT * DS;
+    T* Vb = VT_lds + cur * D * KS;
+    if (kt + 1 < n_kt) load_tile(kt + 1);  // async: in flight during compute
 
-    // ---- S tile = Q K^T (per wave: 16 q x 32 keys) ----
-    floatx4 sc[2];
+    // ---- S tile = Q K^T (16 q x KT keys) ----
+    floatx4 sc[NNT];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
+    for (int nt = 0; nt < NNT; ++nt) {
       sc[nt] = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
-        frag bk = *(const frag*)(K_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        frag bk = *(const frag*)(Kb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
         sc[nt] = MF::mma(q_frag[kc], bk, sc[nt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
-    // ---- mask + online softmax ----
-    float p[2][4];
-    float m_new[4], alpha[4], l_add[4];
+    // ---- mask + online softmax over KT keys ----
+    float p[NNT][4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + hi * 4 + r;
-      float s0 = sc[0][r] * scale, s1 = sc[1][r] * scale;
-      const int k0 = kbase + lo, k1 = kbase + 16 + lo;
-      if (k0 > qrow || k0 >= S) s0 = NEG_BIG;
-      if (k1 > qrow || k1 >= S) s1 = NEG_BIG;
-      float rm = grp16_max(fmaxf(s0, s1));
-      m_new[r] = fmaxf(m_run[r], rm);
-      alpha[r] = __expf(m_run[r] - m_new[r]);
-      p[0][r] = s0 <= NEG_BIG ? 0.f : __expf(s0 - m_new[r]);
-      p[1][r] = s1 <= NEG_BIG ? 0.f : __expf(s1 - m_new[r]);
-      l_add[r] = grp16_sum(p[0][r] + p[1][r]);
-      l_run[r] = l_run[r] * alpha[r] + l_add[r];
-      m_run[r] = m_new[r];
+      float rm = NEG_BIG;
+      float sv[NNT];
+#pragma unroll
+      for (int nt = 0; nt < NNT; ++nt) {
+        const int kk = kbase + nt * 16 + lo;
+        sv[nt] = (kk > qrow || kk >= S) ? NEG_BIG : sc[nt][r] * scale;
+        rm = fmaxf(rm, sv[nt]);
+      }
+      rm = grp16_max(rm);
+      const float m_new = fmaxf(m_run[r], rm);
+      const float alpha = __expf(m_run[r] - m_new);
+      float psum = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < NNT; ++nt) {
+        p[nt][r] = sv[nt] <= NEG_BIG ? 0.f : __expf(sv[nt] - m_new);
+        psum += p[nt][r];
+      }
+      l_run[r] = l_run[r] * alpha + grp16_sum(psum);
+      m_run[r] = m_new;
+#pragma unroll
+      for (int dn = 0; dn < NDN; ++dn) o_acc[dn][r] *= alpha;
     }
-#pragma unroll
-    for (int dn = 0; dn < NDN; ++dn)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[dn][r] *= alpha[r];
 
     // ---- P through per-wave LDS: C-layout -> A-layout ----
     T* Pw = P_lds + wave * 16 * KS;
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt)
+    for (int nt = 0; nt < NNT; ++nt)
 #pragma unroll
       for (int r = 0; r < 4; ++r) Pw[(hi * 4 + r) * KS + nt * 16 + lo] = TR::fromF(p[nt][r]);
-    // same-wave LDS dependency: compiler inserts lgkmcnt waits; no barrier needed
-    frag pa = *(const frag*)(Pw + lo * KS + hi * 8);
+    // same-wave LDS dependency: compiler inserts lgkmcnt waits
+    frag pa[KT / 32];
+#pragma unroll
+    for (int c = 0; c < KT / 32; ++c)
+      pa[c] = *(const frag*)(Pw + lo * KS + c * 32 + hi * 8);
 
     // ---- O += P V ----
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int dn = 0; dn < NDN; ++dn) {
-      frag bv = *(const frag*)(VT_lds + (dn * 16 + lo) * KS + hi * 8);
-      o_acc[dn] = MF::mma(pa, bv, o_acc[dn]);
-    }
+    for (int dn = 0; dn < NDN; ++dn)
+#pragma unroll
+      for (int c = 0; c < KT / 32; ++c) {
+        frag bv = *(const frag*)vt_addr(Vb, dn * 16 + lo, (c * 32 + hi * 8) * 2, KS * 2);
+        o_acc[dn] = MF::mma(pa[c], bv, o_acc[dn]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (kt + 1 < n_kt) write_tile((kt + 1) & 1);  // T14: write late, after compute
+    __syncthreads();
   }
 
   // ---- epilogue ----
@@ -230,8 +284,10 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
 
 // ======================= bwd dK/dV =======================
 // grid over (b, hq, kv-tile of 64 keys); wave owns 16 keys.  Loop q tiles of
-// 32.  LDS: Q[32][D+8] | Q_T[D][32+8] | dO[32][D+8] | dO_T[D][32+8] |
-//          lse[32] f32 | delta[32] f32 | P_T[4][16][32+8]
+// 32 with double-buffered async staging (v2, same structure as forward) and
+// XOR-swizzled transposed images (xmask 3: 64-B data span per row).
+// LDS: Q[2][32][D+8] | Q_T[2][D][32+8] | dO[2][32][D+8] | dO_T[2][D][32+8] |
+//      lse[2][32] f32 | delta[2][32] f32 | P_T[4][16][32+8] | dS[4][16][32+8]
 template <int DT, int D>
 __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     typename DTraits<DT>::T* __restrict__ dk_out,
@@ -253,17 +309,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   constexpr int NDN = D / 16;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  T* Q_lds = (T*)smem_raw;                   // [QT][DS]
-  T* QT_lds = Q_lds + QT * DS;               // [D][QS]
-  T* dO_lds = QT_lds + D * QS;               // [QT][DS]
-  T* dOT_lds = dO_lds + QT * DS;             // [D][QS]
-  T* PT_lds = dOT_lds + D * QS;              // [4][16][QS]  (P^T tiles)
+  T* Q_lds = (T*)smem_raw;                   // [2][QT][DS]
+  T* QT_lds = Q_lds + 2 * QT * DS;           // [2][D][QS] swizzled
+  T* dO_lds = QT_lds + 2 * D * QS;           // [2][QT][DS]
+  T* dOT_lds = dO_lds + 2 * QT * DS;         // [2][D][QS] swizzled
+  T* PT_lds = dOT_lds + 2 * D * QS;          // [4][16][QS]  (P^T tiles)
   T* DS_lds = PT_lds + 4 * 16 * QS;          // [4][16][QS]  (dS^T tiles; separate
                                              //  buffer: avoids an LDS WAR hazard
                                              //  between the P^T A-frag read and
                                              //  the dS^T writes in one iteration)
-  float* lse_lds = (float*)(DS_lds + 4 * 16 * QS);  // [QT]
-  float* dl_lds = lse_lds + QT;                     // [QT]
+  float* lse_lds = (float*)(DS_lds + 4 * 16 * QS);  // [2][QT]
+  float* dl_lds = lse_lds + 2 * QT;                 // [2][QT]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -301,55 +357,82 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   const int qstart = (kt * 64) / QT;         // first q tile that sees these keys
   const int nQT2 = (S + QT - 1) / QT;
 
-  for (int qt = qstart; qt < nQT2; ++qt) {
-    const int qbase = qt * QT;
-    __syncthreads();
-    {  // stage Q, dO (+ transposes), lse, delta
-      constexpr int LPT = (QT * D) / 8;
-      for (int t = threadIdx.x; t < LPT; t += 256) {
-        const int row = t / (D / 8);
-        const int c8 = (t % (D / 8)) * 8;
-        const int qrow = qbase + row;
-        const int qr_c = qrow < S ? qrow : S - 1;
-        frag q8 = *(const frag*)(q + qoff + (int64_t)qr_c * D + c8);
-        *(frag*)(Q_lds + row * DS + c8) = q8;
-        frag d8 = *(const frag*)(do_ + qoff + (int64_t)qr_c * D + c8);
-        *(frag*)(dO_lds + row * DS + c8) = d8;
+  // async staging state (QT*D/8 <= 256 loads: one piece per thread)
+  const int st_t = (int)threadIdx.x;
+  const bool st_on = st_t < (QT * D) / 8;
+  const int st_row = st_t / (D / 8);
+  const int st_c8 = (st_t % (D / 8)) * 8;
+  frag qreg, dreg;
+  float lse_reg = 0.f, dl_reg = 0.f;
+
+  auto load_qtile = [&](int qt) {
+    const int qrow = qt * QT + st_row;
+    const int qr_c = qrow < S ? qrow : S - 1;
+    if (st_on) {
+      qreg = *(const frag*)(q + qoff + (int64_t)qr_c * D + st_c8);
+      dreg = *(const frag*)(do_ + qoff + (int64_t)qr_c * D + st_c8);
+    }
+    if (st_t < QT) {
+      const int rr = qt * QT + st_t;
+      const int rr_c = rr < S ? rr : S - 1;
+      lse_reg = lse[lseoff + rr_c];
+      dl_reg = delta[lseoff + rr_c];
+    }
+  };
+  auto write_qtile = [&](int buf) {
+    if (st_on) {
+      *(frag*)(Q_lds + buf * QT * DS + st_row * DS + st_c8) = qreg;
+      *(frag*)(dO_lds + buf * QT * DS + st_row * DS + st_c8) = dreg;
+      T* QTb = QT_lds + buf * D * QS;
+      T* dOTb = dOT_lds + buf * D * QS;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          QT_lds[(c8 + j) * QS + row] = ((const T*)&q8)[j];
-          dOT_lds[(c8 + j) * QS + row] = ((const T*)&d8)[j];
-        }
-      }
-      for (int t = threadIdx.x; t < QT; t += 256) {
-        const int qrow = qbase + t;
-        const int qr_c = qrow < S ? qrow : S - 1;
-        lse_lds[t] = lse[lseoff + qr_c];
-        dl_lds[t] = delta[lseoff + qr_c];
+      for (int j = 0; j < 8; ++j) {
+        *vt_addr(QTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&qreg)[j];
+        *vt_addr(dOTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&dreg)[j];
       }
     }
-    __syncthreads();
+    if (st_t < QT) {
+      lse_lds[buf * QT + st_t] = lse_reg;
+      dl_lds[buf * QT + st_t] = dl_reg;
+    }
+  };
+
+  load_qtile(qstart);
+  write_qtile(0);
+  __syncthreads();
+
+  for (int qt = qstart; qt < nQT2; ++qt) {
+    const int qbase = qt * QT;
+    const int cur = (qt - qstart) & 1;
+    T* Qb = Q_lds + cur * QT * DS;
+    T* dOb = dO_lds + cur * QT * DS;
+    T* QTb = QT_lds + cur * D * QS;
+    T* dOTb = dOT_lds + cur * D * QS;
+    const float* lse_b = lse_lds + cur * QT;
+    const float* dl_b = dl_lds + cur * QT;
+    if (qt + 1 < nQT2) load_qtile(qt + 1);
 
     // ---- S^T = K Q^T (16 keys x 32 q), P^T = exp(scale*S^T - lse) ----
     float pt[2][4], dst[2][4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
       floatx4 st = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
-        frag bq = *(const frag*)(Q_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        frag bq = *(const frag*)(Qb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
         st = MF::mma(k_frag[kc], bq, st);
       }
       // dP^T = V dO^T
       floatx4 dpt = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
-        frag bd = *(const frag*)(dO_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        frag bd = *(const frag*)(dOb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
         dpt = MF::mma(v_frag[kc], bd, dpt);
       }
       const int qcol = qbase + nt * 16 + lo;
-      const float lse_q = lse_lds[nt * 16 + lo];
-      const float dl_q = dl_lds[nt * 16 + lo];
+      const float lse_q = lse_b[nt * 16 + lo];
+      const float dl_q = dl_b[nt * 16 + lo];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int krow = k0 + hi * 4 + r;
@@ -370,7 +453,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     frag pa = *(const frag*)(Pw + lo * QS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
-      frag bd = *(const frag*)(dOT_lds + (dn * 16 + lo) * QS + hi * 8);
+      frag bd = *(const frag*)vt_addr(dOTb, dn * 16 + lo, hi * 8 * 2, QS * 2, 3);
       dv_acc[dn] = MF::mma(pa, bd, dv_acc[dn]);
     }
 
@@ -383,9 +466,13 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     frag da = *(const frag*)(Dw + lo * QS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
-      frag bq = *(const frag*)(QT_lds + (dn * 16 + lo) * QS + hi * 8);
+      frag bq = *(const frag*)vt_addr(QTb, dn * 16 + lo, hi * 8 * 2, QS * 2, 3);
       dk_acc[dn] = MF::mma(da, bq, dk_acc[dn]);
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (qt + 1 < nQT2) write_qtile(cur ^ 1);  // T14: write late
+    __syncthreads();
   }
 
   // ---- write dK, dV (per q-head layout [B,Hq,S,D]; caller sums GQA groups) ----
@@ -424,10 +511,10 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   constexpr int NDN = D / 16;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  T* K_lds = (T*)smem_raw;                    // [KT][DS]
-  T* V_lds = K_lds + KT * DS;                 // [KT][DS]
-  T* KT_lds = V_lds + KT * DS;                // [D][KS]
-  T* S_lds = KT_lds + D * KS;                 // [4][16][KS]
+  T* K_lds = (T*)smem_raw;                    // [2][KT][DS]
+  T* V_lds = K_lds + 2 * KT * DS;             // [2][KT][DS]
+  T* KT_lds = V_lds + 2 * KT * DS;            // [2][D][KS] swizzled (xmask 3)
+  T* S_lds = KT_lds + 2 * D * KS;             // [4][16][KS]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -472,35 +559,54 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int kv_end = min(S, qt * 64 + 64);
   const int n_kt = (kv_end + KT - 1) / KT;
 
+  // async double-buffered staging (one piece per thread: KT*D/8 <= 256)
+  const int st_t = (int)threadIdx.x;
+  const bool st_on = st_t < (KT * D) / 8;
+  const int st_row = st_t / (D / 8);
+  const int st_c8 = (st_t % (D / 8)) * 8;
+  frag kreg, vreg;
+
+  auto load_ktile = [&](int kt) {
+    if (st_on) {
+      const int krow = kt * KT + st_row;
+      const int kr_c = krow < S ? krow : S - 1;
+      kreg = *(const frag*)(k + kvoff + (int64_t)kr_c * D + st_c8);
+      vreg = *(const frag*)(v + kvoff + (int64_t)kr_c * D + st_c8);
+    }
+  };
+  auto write_ktile = [&](int buf) {
+    if (st_on) {
+      *(frag*)(K_lds + buf * KT * DS + st_row * DS + st_c8) = kreg;
+      *(frag*)(V_lds + buf * KT * DS + st_row * DS + st_c8) = vreg;
+      T* KTb = KT_lds + buf * D * KS;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *vt_addr(KTb, st_c8 + j, st_row * 2, KS * 2, 3) = ((const T*)&kreg)[j];
+    }
+  };
+
+  load_ktile(0);
+  write_ktile(0);
+  __syncthreads();
+
   for (int kt = 0; kt < n_kt; ++kt) {
     const int kbase = kt * KT;
-    __syncthreads();
-    {
-      constexpr int LPT = (KT * D) / 8;
-      for (int t = threadIdx.x; t < LPT; t += 256) {
-        const int row = t / (D / 8);
-        const int c8 = (t % (D / 8)) * 8;
-        const int krow = kbase + row;
-        const int kr_c = krow < S ? krow : S - 1;
-        frag k8 = *(const frag*)(k + kvoff + (int64_t)kr_c * D + c8);
-        *(frag*)(K_lds + row * DS + c8) = k8;
-        frag v8 = *(const frag*)(v + kvoff + (int64_t)kr_c * D + c8);
-        *(frag*)(V_lds + row * DS + c8) = v8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) KT_lds[(c8 + j) * KS + row] = ((const T*)&k8)[j];
-      }
-    }
-    __syncthreads();
+    const int cur = kt & 1;
+    T* Kb = K_lds + cur * KT * DS;
+    T* Vb = V_lds + cur * KT * DS;
+    T* KTb = KT_lds + cur * D * KS;
+    if (kt + 1 < n_kt) load_ktile(kt + 1);
 
     float ds[2][4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
       floatx4 st = (floatx4)(0.f), dpt = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
-        frag bk = *(const frag*)(K_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        frag bk = *(const frag*)(Kb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
         st = MF::mma(q_frag[kc], bk, st);
-        frag bv = *(const frag*)(V_lds + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        frag bv = *(const frag*)(Vb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
         dpt = MF::mma(do_frag[kc], bv, dpt);
       }
       const int kcol = kbase + nt * 16 + lo;
@@ -523,9 +629,13 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     frag da = *(const frag*)(Sw + lo * KS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
-      frag bk = *(const frag*)(KT_lds + (dn * 16 + lo) * KS + hi * 8);
+      frag bk = *(const frag*)vt_addr(KTb, dn * 16 + lo, hi * 8 * 2, KS * 2, 3);
       dq_acc[dn] = MF::mma(da, bk, dq_acc[dn]);
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (kt + 1 < n_kt) write_ktile(cur ^ 1);  // T14: write late
+    __syncthreads();
   }
 
 #pragma unroll
@@ -547,8 +657,8 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
   using T = typename DTraits<DT>::T;
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
-  constexpr int KT = 32, KS = KT + 8, DS = D + 8;
-  const size_t lds = sizeof(T) * (KT * DS + D * KS + 4 * 16 * KS);
+  constexpr int KT = 64, KS = KT + 8, DS = D + 8;
+  const size_t lds = sizeof(T) * (2 * KT * DS + 2 * D * KS + 4 * 16 * KS);
   hipLaunchKernelGGL((attn_fwd_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
                      (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale);
@@ -599,7 +709,7 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
   const int nKT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nKT);
   constexpr int QT = 32, QS = QT + 8, DS = D + 8;
-  const size_t lds = sizeof(T) * (2 * QT * DS + 2 * D * QS + 2 * 4 * 16 * QS) + sizeof(float) * 2 * QT;
+  const size_t lds = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 4 * 16 * QS) + sizeof(float) * 4 * QT;
   hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
                      (const T*)q, (const T*)k, (const T*)v, lse, delta,
@@ -633,7 +743,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
   constexpr int KT = 32, KS = KT + 8, DS = D + 8;
-  const size_t lds = sizeof(T) * (2 * KT * DS + D * KS + 4 * 16 * KS);
+  const size_t lds = sizeof(T) * (4 * KT * DS + 2 * D * KS + 4 * 16 * KS);
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
