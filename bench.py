@@ -79,6 +79,10 @@ def main():
     device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0)))
     torch.cuda.set_device(device)
 
+    from opendiloco_amd.gemm_tuning import enable_tuned_gemms
+
+    enable_tuned_gemms()
+
     from opendiloco_amd.diloco import DiLoCoOptimizer
     from opendiloco_amd.model import LlamaForCausalLM
     from opendiloco_amd.optim import clip_grad_norm_flat_

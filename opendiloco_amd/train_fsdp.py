@@ -133,6 +133,10 @@ def train(config: Config):
         seed=config.data_seed, rank=rank)
 
     # ---- model ----
+    if device.type == "cuda":
+        from opendiloco_amd.gemm_tuning import enable_tuned_gemms
+
+        enable_tuned_gemms()
     model = LlamaForCausalLM.from_pretrained(config.path_model)
     model = model.to(device)
     model.compute_dtype = _compute_dtype(config.precision, device)
