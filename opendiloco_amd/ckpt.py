@@ -104,6 +104,9 @@ def load_checkpoint(checkpoint_path: str, model, optimizer, scheduler=None,
         outer_optimizer.load_state_dict(global_state["outer_optimizer"])
     if scaler is not None:
         scaler.load_state_dict(global_state["scaler"])
+    from opendiloco_amd.model import bump_weights_version
+
+    bump_weights_version()  # master weights changed in place
     return global_state["loss"]
 
 

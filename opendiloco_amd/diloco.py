@@ -344,6 +344,9 @@ class DiLoCoOptimizer:
         hivemind state download, train_fsdp.py:348-349)."""
         if _world_size() <= 1:
             return
+        from opendiloco_amd.model import bump_weights_version
+
+        bump_weights_version()
         for t in (self.flat.flat_param, self.inner_optimizer.flat_m,
                   self.inner_optimizer.flat_v, self.state_averager.flat_outer):
             dist.broadcast(t, src=0)

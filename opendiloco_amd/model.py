@@ -42,18 +42,58 @@ def _cast(w: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     return w if w.dtype == dtype else w.to(dtype)
 
 
+# version counter for cached low-precision weight copies: bumped by whatever
+# mutates master weights in place (FusedAdamW.step, the outer step, loads)
+_WEIGHTS_VERSION = [0]
+
+
+def bump_weights_version() -> None:
+    _WEIGHTS_VERSION[0] += 1
+
+
+class _CastLinearFn(torch.autograd.Function):
+    """Linear over a cached low-precision weight copy, with the weight grad
+    routed to the fp32 master (autocast semantics; the cache saves the
+    fp32->bf16 cast per microbatch that the per-forward `.to()` would pay)."""
+
+    @staticmethod
+    def forward(ctx, x, w_master, w16):
+        ctx.save_for_backward(x, w16)
+        return F.linear(x, w16)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w16 = ctx.saved_tensors
+        dx = dy @ w16
+        dw = (dy.reshape(-1, dy.shape[-1]).t() @ x.reshape(-1, x.shape[-1]))
+        return dx, dw.to(torch.float32), None
+
+
 class CastLinear(nn.Module):
-    """Bias-free linear with an fp32 master weight cast to compute dtype per
-    forward (autocast semantics); the GEMM itself is a rocBLAS bf16 GEMM with
+    """Bias-free linear with an fp32 master weight cast to compute dtype
+    (cached per optimizer step); the GEMM itself is a rocBLAS bf16 GEMM with
     fp32 accumulate — a plain library GEMM, per the MFMA design rules."""
 
     def __init__(self, in_features: int, out_features: int):
         super().__init__()
         self.weight = nn.Parameter(torch.empty(out_features, in_features))
         self.in_features, self.out_features = in_features, out_features
+        self._w16 = None
+        self._ver = -1
+
+    def _cached_w16(self, dtype: torch.dtype) -> torch.Tensor:
+        if self._ver != _WEIGHTS_VERSION[0] or self._w16 is None or self._w16.dtype != dtype:
+            with torch.no_grad():
+                self._w16 = self.weight.to(dtype)
+            self._ver = _WEIGHTS_VERSION[0]
+        return self._w16
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, _cast(self.weight, x.dtype))
+        if x.dtype == self.weight.dtype:
+            return F.linear(x, self.weight)
+        if not x.is_cuda:
+            return F.linear(x, _cast(self.weight, x.dtype))
+        return _CastLinearFn.apply(x, self.weight, self._cached_w16(x.dtype))
 
 
 class RMSNorm(nn.Module):
@@ -196,6 +236,16 @@ class LlamaForCausalLM(nn.Module):
             model.init_weights()
         return model
 
+    def _cached_embed(self, cdtype: torch.dtype) -> torch.Tensor:
+        w = self.model.embed_tokens.weight
+        if w.dtype == cdtype or not w.is_cuda:
+            return _cast(w, cdtype)
+        if getattr(self, "_embed_ver", -1) != _WEIGHTS_VERSION[0] or self._embed_w16.dtype != cdtype:
+            with torch.no_grad():
+                self._embed_w16 = w.to(cdtype)
+            self._embed_ver = _WEIGHTS_VERSION[0]
+        return self._embed_w16
+
     def _dtype_for(self, device: torch.device) -> torch.dtype:
         if self.compute_dtype is not None:
             return self.compute_dtype
@@ -210,7 +260,7 @@ class LlamaForCausalLM(nn.Module):
         S = input_ids.shape[1]
         assert S <= self.config.max_position_embeddings
 
-        h = F.embedding(input_ids, _cast(self.model.embed_tokens.weight, cdtype))
+        h = F.embedding(input_ids, self._cached_embed(cdtype))
         for layer in self.model.layers:
             h = layer(h, self.rotary)
         h = self.model.norm(h)

@@ -110,6 +110,9 @@ class FusedAdamW(torch.optim.Optimizer):
         t = self._step_count_t
         for st in self.state.values():
             st["step"] += 1
+        from opendiloco_amd.model import bump_weights_version
+
+        bump_weights_version()  # invalidate cached bf16 weight copies
         if self.flat.device.type == "cuda":
             from opendiloco_amd.ops import _ext
 
@@ -171,6 +174,9 @@ class FlatSGDNesterov:
 
     def step_fused(self, flat_local: torch.Tensor, g_avg: torch.Tensor):
         """theta_outer/momentum update + theta_local copy-back."""
+        from opendiloco_amd.model import bump_weights_version
+
+        bump_weights_version()
         self.lr = self.param_groups[0]["lr"]
         first = self.momentum_buf is None
         if first:
