@@ -195,12 +195,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + online softmax over KT keys ----
-    // full tiles (strictly below the diagonal, fully in-bounds) skip the
-    // per-element causal/boundary masking entirely
-    const bool full_tile = (kbase + KT <= q0) && (kbase + KT <= S);
     float p[NNT][4];
-    float alpha[4];
-    bool any_rescale = false;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + hi * 4 + r;
@@ -208,32 +203,23 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       float sv[NNT];
 #pragma unroll
       for (int nt = 0; nt < NNT; ++nt) {
-        if (full_tile) {
-          sv[nt] = sc[nt][r] * scale;
-        } else {
-          const int kk = kbase + nt * 16 + lo;
-          sv[nt] = (kk > qrow || kk >= S) ? NEG_BIG : sc[nt][r] * scale;
-        }
+        const int kk = kbase + nt * 16 + lo;
+        sv[nt] = (kk > qrow || kk >= S) ? NEG_BIG : sc[nt][r] * scale;
         rm = fmaxf(rm, sv[nt]);
       }
       rm = grp16_max(rm);
       const float m_new = fmaxf(m_run[r], rm);
-      alpha[r] = (m_new == m_run[r]) ? 1.f : __expf(m_run[r] - m_new);
-      any_rescale |= (m_new != m_run[r]);
+      const float alpha = __expf(m_run[r] - m_new);
       float psum = 0.f;
 #pragma unroll
       for (int nt = 0; nt < NNT; ++nt) {
         p[nt][r] = sv[nt] <= NEG_BIG ? 0.f : __expf(sv[nt] - m_new);
         psum += p[nt][r];
       }
-      l_run[r] = l_run[r] * alpha[r] + grp16_sum(psum);
+      l_run[r] = l_run[r] * alpha + grp16_sum(psum);
       m_run[r] = m_new;
-    }
-    if (__any(any_rescale)) {
 #pragma unroll
-      for (int dn = 0; dn < NDN; ++dn)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) o_acc[dn][r] *= alpha[r];
+      for (int dn = 0; dn < NDN; ++dn) o_acc[dn][r] *= alpha;
     }
 
     // ---- P through per-wave LDS: C-layout -> A-layout ----
@@ -447,20 +433,12 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
       const int qcol = qbase + nt * 16 + lo;
       const float lse_q = lse_b[nt * 16 + lo];
       const float dl_q = dl_b[nt * 16 + lo];
-      // full tile: every q in this sub-tile is >= every key of the wave and
-      // in bounds -> no per-element predicate
-      const bool ft = (qbase + nt * 16 >= k0 + 16) && (qbase + nt * 16 + 16 <= S);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float pv;
-        if (ft) {
+        const int krow = k0 + hi * 4 + r;
+        float pv = 0.f;
+        if (krow <= qcol && krow < S && qcol < S)
           pv = __expf(st[r] * scale - lse_q);
-        } else {
-          const int krow = k0 + hi * 4 + r;
-          pv = 0.f;
-          if (krow <= qcol && krow < S && qcol < S)
-            pv = __expf(st[r] * scale - lse_q);
-        }
         pt[nt][r] = pv;
         dst[nt][r] = pv * (dpt[r] - dl_q) * scale;
       }
@@ -632,19 +610,12 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         dpt = MF::mma(do_frag[kc], bv, dpt);
       }
       const int kcol = kbase + nt * 16 + lo;
-      const bool ft = (kbase + nt * 16 + 16 <= q0) && (kbase + nt * 16 + 16 <= S)
-                      && (q0 + 16 <= S);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float pv;
-        if (ft) {
+        const int qrow = q0 + hi * 4 + r;
+        float pv = 0.f;
+        if (kcol <= qrow && kcol < S && qrow < S)
           pv = __expf(st[r] * scale - lse_r[r]);
-        } else {
-          const int qrow = q0 + hi * 4 + r;
-          pv = 0.f;
-          if (kcol <= qrow && kcol < S && qrow < S)
-            pv = __expf(st[r] * scale - lse_r[r]);
-        }
         ds[nt][r] = pv * (dpt[r] - dl_r[r]) * scale;
       }
     }

@@ -69,6 +69,25 @@ class _CastLinearFn(torch.autograd.Function):
         return dx, dw.to(torch.float32), None
 
 
+class _CastEmbeddingFn(torch.autograd.Function):
+    """Embedding lookup over a cached low-precision weight copy; the weight
+    grad goes to the fp32 master via the same aten dense backward the
+    uncached `.to()` path used (deterministic, matches previous numerics)."""
+
+    @staticmethod
+    def forward(ctx, ids, w_master, w16):
+        ctx.save_for_backward(ids)
+        ctx.num_weights = w_master.shape[0]
+        return F.embedding(ids, w16)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (ids,) = ctx.saved_tensors
+        dw = torch.ops.aten.embedding_dense_backward(
+            dy, ids, ctx.num_weights, -1, False)
+        return None, dw.to(torch.float32), None
+
+
 class CastLinear(nn.Module):
     """Bias-free linear with an fp32 master weight cast to compute dtype
     (cached per optimizer step); the GEMM itself is a rocBLAS bf16 GEMM with
@@ -238,11 +257,9 @@ class LlamaForCausalLM(nn.Module):
 
     def _cached_embed(self, cdtype: torch.dtype) -> torch.Tensor:
         w = self.model.embed_tokens.weight
-        if w.dtype == cdtype or not w.is_cuda:
-            return _cast(w, cdtype)
-        if getattr(self, "_embed_ver", -1) != _WEIGHTS_VERSION[0] or self._embed_w16.dtype != cdtype:
+        if getattr(self, "_embed_ver", -1) != _WEIGHTS_VERSION[0]                 or getattr(self, "_embed_w16", None) is None or self._embed_w16.dtype != cdtype:
             with torch.no_grad():
-                self._embed_w16 = w.to(cdtype)
+                self._embed_w16 = w.to(cdtype).contiguous()
             self._embed_ver = _WEIGHTS_VERSION[0]
         return self._embed_w16
 
@@ -260,7 +277,11 @@ class LlamaForCausalLM(nn.Module):
         S = input_ids.shape[1]
         assert S <= self.config.max_position_embeddings
 
-        h = F.embedding(input_ids, self._cached_embed(cdtype))
+        if cdtype == self.model.embed_tokens.weight.dtype or not input_ids.is_cuda:
+            h = F.embedding(input_ids, _cast(self.model.embed_tokens.weight, cdtype))
+        else:
+            h = _CastEmbeddingFn.apply(input_ids, self.model.embed_tokens.weight,
+                                       self._cached_embed(cdtype))
         for layer in self.model.layers:
             h = layer(h, self.rotary)
         h = self.model.norm(h)
