@@ -173,14 +173,22 @@ def main():
         dur_s = ev0.elapsed_time(ev1) / 1000.0 / reps
         algo_bytes = 28.0 * f.n
         peak = 8.0e12  # HBM3E spec peak (MI355X_MICROARCH.md; ~6.3e12 achievable)
-        traffic = os.environ.get("DK_ADAMW_TRAFFIC_BYTES")
+        # measured per-launch HBM traffic from the committed PMC run
+        # (profiles/round1_adamw_traffic.json: FETCH_SIZE x2-corrected + WRITE_SIZE,
+        # measured at the llama-150m size; scaled by n for other sizes)
+        traffic = None
+        tf_path = os.path.join(REPO_ROOT, "profiles", "round1_adamw_traffic.json")
+        if os.path.exists(tf_path):
+            with open(tf_path) as tf:
+                tj = json.load(tf)
+            traffic = tj["bytes_per_param"] * f.n
         roofline = {
             "bound": "hbm",
             "achieved": algo_bytes / dur_s / 1e9,
             "peak": peak / 1e9,
             "unit": "GB/s",
             "frac": (algo_bytes / dur_s) / peak,
-            "traffic": float(traffic) if traffic else None,
+            "traffic": traffic,
             "kernel": "dk_fused_adamw",
             "launch_ms": dur_s * 1000.0,
         }

@@ -195,17 +195,28 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + online softmax over KT keys ----
+    // tiles strictly below the diagonal and fully in-bounds need no
+    // per-element mask (wave-uniform branch; saves 2 cmp+sel per element)
+    const bool full_tile = (kbase + KT <= q0) && (kbase + KT <= S);
     float p[NNT][4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + hi * 4 + r;
       float rm = NEG_BIG;
       float sv[NNT];
+      if (full_tile) {
 #pragma unroll
-      for (int nt = 0; nt < NNT; ++nt) {
-        const int kk = kbase + nt * 16 + lo;
-        sv[nt] = (kk > qrow || kk >= S) ? NEG_BIG : sc[nt][r] * scale;
-        rm = fmaxf(rm, sv[nt]);
+        for (int nt = 0; nt < NNT; ++nt) {
+          sv[nt] = sc[nt][r] * scale;
+          rm = fmaxf(rm, sv[nt]);
+        }
+      } else {
+#pragma unroll
+        for (int nt = 0; nt < NNT; ++nt) {
+          const int kk = kbase + nt * 16 + lo;
+          sv[nt] = (kk > qrow || kk >= S) ? NEG_BIG : sc[nt][r] * scale;
+          rm = fmaxf(rm, sv[nt]);
+        }
       }
       rm = grp16_max(rm);
       const float m_new = fmaxf(m_run[r], rm);
