@@ -64,6 +64,12 @@ int dk_swiglu_fwd(void* y, const void* gate, const void* up, int64_t n,
                   int dtype, dkStream stream);
 int dk_swiglu_bwd(void* dgate, void* dup, const void* dy, const void* gate,
                   const void* up, int64_t n, int dtype, dkStream stream);
+/* fused-layout variant over the batched gate-up GEMM output gu [rows, 2*I]:
+ * y[r,c] = silu(gu[r,c]) * gu[r,I+c]; bwd writes dgu in the same layout. */
+int dk_swiglu2_fwd(void* y, const void* gu, int64_t rows, int64_t I, int dtype,
+                   dkStream stream);
+int dk_swiglu2_bwd(void* dgu, const void* dy, const void* gu, int64_t rows,
+                   int64_t I, int dtype, dkStream stream);
 
 /* ---- Fused cross-entropy ------------------------------------------------
  * Replaces the causal-LM loss of LlamaForCausalLM (logits->fp32, shifted CE,

@@ -98,6 +98,24 @@ std::vector<at::Tensor> swiglu_bwd(const at::Tensor& dy, const at::Tensor& gate,
   return {dgate, dup};
 }
 
+at::Tensor swiglu2_fwd(const at::Tensor& gu, int64_t I) {
+  CHECK_DEV_CONTIG(gu);
+  const int64_t rows = gu.numel() / (2 * I);
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto y = at::empty(sizes, gu.options());
+  DK_OK(dk_swiglu2_fwd(y.data_ptr(), gu.data_ptr(), rows, I, dt_of(gu), stream()));
+  return y;
+}
+
+at::Tensor swiglu2_bwd(const at::Tensor& dy, const at::Tensor& gu, int64_t I) {
+  auto dgu = at::empty_like(gu);
+  const int64_t rows = gu.numel() / (2 * I);
+  DK_OK(dk_swiglu2_bwd(dgu.data_ptr(), dy.data_ptr(), gu.data_ptr(), rows, I,
+                       dt_of(gu), stream()));
+  return dgu;
+}
+
 // ---- cross entropy ----
 std::vector<at::Tensor> ce_fwd(const at::Tensor& logits, const at::Tensor& labels) {
   CHECK_DEV_CONTIG(logits);
@@ -237,6 +255,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope", &rope);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("swiglu2_fwd", &swiglu2_fwd);
+  m.def("swiglu2_bwd", &swiglu2_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("attn_fwd", &attn_fwd);

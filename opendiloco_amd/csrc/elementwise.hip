@@ -354,6 +354,98 @@ __global__ void swiglu_bwd_kernel(typename DTraits<DT>::T* __restrict__ dgate,
   }
 }
 
+// fused-layout variant: gate/up live interleaved per row in one [R, 2*I]
+// tensor (the batched gate-up GEMM's output); avoids the split+contiguous
+// copies entirely.  y[r,c] = silu(gu[r,c]) * gu[r,I+c].
+template <int DT>
+__global__ void swiglu2_fwd_kernel(typename DTraits<DT>::T* __restrict__ y,
+                                   const typename DTraits<DT>::T* __restrict__ gu,
+                                   int64_t rows, int64_t nvec_per_row, int64_t I) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  constexpr int W = VecIO<T>::W;
+  using V = typename VecIO<T>::V;
+  const int64_t total = rows * nvec_per_row;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = idx / nvec_per_row;
+    const int64_t c = (idx % nvec_per_row) * W;
+    V gv = *(const V*)(gu + r * 2 * I + c);
+    V uv = *(const V*)(gu + r * 2 * I + I + c);
+    V yv;
+#pragma unroll
+    for (int j = 0; j < W; ++j) {
+      float g = TR::toF(((const T*)&gv)[j]);
+      float u = TR::toF(((const T*)&uv)[j]);
+      float sig = 1.f / (1.f + __expf(-g));
+      ((T*)&yv)[j] = TR::fromF(g * sig * u);
+    }
+    *(V*)(y + r * I + c) = yv;
+  }
+}
+
+template <int DT>
+__global__ void swiglu2_bwd_kernel(typename DTraits<DT>::T* __restrict__ dgu,
+                                   const typename DTraits<DT>::T* __restrict__ dy,
+                                   const typename DTraits<DT>::T* __restrict__ gu,
+                                   int64_t rows, int64_t nvec_per_row, int64_t I) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  constexpr int W = VecIO<T>::W;
+  using V = typename VecIO<T>::V;
+  const int64_t total = rows * nvec_per_row;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t r = idx / nvec_per_row;
+    const int64_t c = (idx % nvec_per_row) * W;
+    V gv = *(const V*)(gu + r * 2 * I + c);
+    V uv = *(const V*)(gu + r * 2 * I + I + c);
+    V dyv = *(const V*)(dy + r * I + c);
+    V dgv, duv;
+#pragma unroll
+    for (int j = 0; j < W; ++j) {
+      float g = TR::toF(((const T*)&gv)[j]);
+      float u = TR::toF(((const T*)&uv)[j]);
+      float d = TR::toF(((const T*)&dyv)[j]);
+      float sig = 1.f / (1.f + __expf(-g));
+      float silu = g * sig;
+      float dsilu = sig * (1.f + g * (1.f - sig));
+      ((T*)&dgv)[j] = TR::fromF(d * u * dsilu);
+      ((T*)&duv)[j] = TR::fromF(d * silu);
+    }
+    *(V*)(dgu + r * 2 * I + c) = dgv;
+    *(V*)(dgu + r * 2 * I + I + c) = duv;
+  }
+}
+
+extern "C" int dk_swiglu2_fwd(void* y, const void* gu, int64_t rows, int64_t I,
+                              int dtype, dkStream stream) {
+  DK_DISPATCH_DT(dtype, {
+    using T = typename DTraits<kDT>::T;
+    constexpr int W = VecIO<T>::W;
+    if (I % W) return (int)hipErrorInvalidValue;
+    int grid = dk_stream_grid(rows * (I / W), 256);
+    hipLaunchKernelGGL((swiglu2_fwd_kernel<kDT>), dim3(grid), dim3(256), 0,
+                       (hipStream_t)stream, (T*)y, (const T*)gu, rows, I / W, I);
+  });
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int dk_swiglu2_bwd(void* dgu, const void* dy, const void* gu, int64_t rows,
+                              int64_t I, int dtype, dkStream stream) {
+  DK_DISPATCH_DT(dtype, {
+    using T = typename DTraits<kDT>::T;
+    constexpr int W = VecIO<T>::W;
+    if (I % W) return (int)hipErrorInvalidValue;
+    int grid = dk_stream_grid(rows * (I / W), 256);
+    hipLaunchKernelGGL((swiglu2_bwd_kernel<kDT>), dim3(grid), dim3(256), 0,
+                       (hipStream_t)stream, (T*)dgu, (const T*)dy, (const T*)gu, rows, I / W, I);
+  });
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
 extern "C" int dk_swiglu_fwd(void* y, const void* gate, const void* up, int64_t n,
                              int dtype, dkStream stream) {
   DK_DISPATCH_DT(dtype, {

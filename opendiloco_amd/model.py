@@ -115,6 +115,51 @@ class CastLinear(nn.Module):
         return _CastLinearFn.apply(x, self.weight, self._cached_w16(x.dtype))
 
 
+class _MultiCastLinearFn(torch.autograd.Function):
+    """One GEMM over a cached concatenation of several bf16 weight copies
+    (QKV / gate-up batching): fewer, larger rocBLAS GEMMs; weight grads are
+    sliced back to the separate fp32 masters (state_dict layout unchanged)."""
+
+    @staticmethod
+    def forward(ctx, x, w16cat, sizes, *masters):
+        ctx.save_for_backward(x, w16cat)
+        ctx.sizes = sizes
+        return F.linear(x, w16cat)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w16cat = ctx.saved_tensors
+        dx = dy @ w16cat
+        dwcat = dy.reshape(-1, dy.shape[-1]).t() @ x.reshape(-1, x.shape[-1])
+        dws = []
+        off = 0
+        for n in ctx.sizes:
+            dws.append(dwcat[off:off + n].to(torch.float32))
+            off += n
+        return (dx, None, None, *dws)
+
+
+class FusedProj:
+    """Cache of a concatenated low-precision copy of several CastLinear
+    weights, invalidated by the weights version counter."""
+
+    def __init__(self, linears: list[CastLinear]):
+        self.linears = linears
+        self.sizes = [l.out_features for l in linears]
+        self._w16 = None
+        self._ver = -1
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        if not x.is_cuda or x.dtype == self.linears[0].weight.dtype:
+            return torch.cat([l(x) for l in self.linears], dim=-1)
+        if self._ver != _WEIGHTS_VERSION[0] or self._w16 is None or self._w16.dtype != x.dtype:
+            with torch.no_grad():
+                self._w16 = torch.cat([l.weight.to(x.dtype) for l in self.linears], dim=0)
+            self._ver = _WEIGHTS_VERSION[0]
+        return _MultiCastLinearFn.apply(x, self._w16, self.sizes,
+                                        *[l.weight for l in self.linears])
+
+
 class RMSNorm(nn.Module):
     def __init__(self, hidden: int, eps: float):
         super().__init__()
@@ -154,13 +199,17 @@ class Attention(nn.Module):
         self.v_proj = CastLinear(h, self.num_kv_heads * D)
         self.o_proj = CastLinear(self.num_heads * D, h)
         self.scale = D ** -0.5
+        self._qkv = FusedProj([self.q_proj, self.k_proj, self.v_proj])
 
     def forward(self, x: torch.Tensor, rotary: Rotary) -> torch.Tensor:
         B, S, _ = x.shape
         D = self.head_dim
-        q = self.q_proj(x).view(B, S, self.num_heads, D).transpose(1, 2).contiguous()
-        k = self.k_proj(x).view(B, S, self.num_kv_heads, D).transpose(1, 2).contiguous()
-        v = self.v_proj(x).view(B, S, self.num_kv_heads, D).transpose(1, 2).contiguous()
+        qkv = self._qkv(x)
+        nq, nkv = self.num_heads * D, self.num_kv_heads * D
+        q, k, v = qkv.split([nq, nkv, nkv], dim=-1)
+        q = q.view(B, S, self.num_heads, D).transpose(1, 2).contiguous()
+        k = k.view(B, S, self.num_kv_heads, D).transpose(1, 2).contiguous()
+        v = v.view(B, S, self.num_kv_heads, D).transpose(1, 2).contiguous()
         q = ops.rope(q, rotary.cos, rotary.sin, S)
         k = ops.rope(k, rotary.cos, rotary.sin, S)
         o = ops.attention(q, k, v, self.scale)  # [B, Hq, S, D]
@@ -175,8 +224,12 @@ class MLP(nn.Module):
         self.gate_proj = CastLinear(h, i)
         self.up_proj = CastLinear(h, i)
         self.down_proj = CastLinear(i, h)
+        self._gu = FusedProj([self.gate_proj, self.up_proj])
+        self._inter = i
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda:
+            return self.down_proj(ops.swiglu_fused(self._gu(x), self._inter))
         return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
 
 

@@ -156,6 +156,36 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return _SwiGLUFn.apply(gate, up)
 
 
+class _SwiGLUFusedFn(torch.autograd.Function):
+    """SwiGLU over the batched gate-up GEMM output gu [.., 2*I]
+    (gate = gu[..., :I], up = gu[..., I:]) — avoids split+contiguous copies."""
+
+    @staticmethod
+    def forward(ctx, gu, inter):
+        ctx.save_for_backward(gu)
+        ctx.inter = inter
+        if gu.is_cuda:
+            return _ext().swiglu2_fwd(gu.contiguous(), inter)
+        g, u = gu[..., :inter].float(), gu[..., inter:].float()
+        return (torch.nn.functional.silu(g) * u).to(gu.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (gu,) = ctx.saved_tensors
+        inter = ctx.inter
+        if gu.is_cuda:
+            return _ext().swiglu2_bwd(dy.contiguous(), gu, inter), None
+        g, u, d = gu[..., :inter].float(), gu[..., inter:].float(), dy.float()
+        sig = torch.sigmoid(g)
+        dgate = d * u * (sig * (1 + g * (1 - sig)))
+        dup = d * g * sig
+        return torch.cat([dgate, dup], dim=-1).to(gu.dtype), None
+
+
+def swiglu_fused(gu: torch.Tensor, inter: int) -> torch.Tensor:
+    return _SwiGLUFusedFn.apply(gu, inter)
+
+
 # ====================== causal-LM cross entropy ======================
 # transformers LlamaForCausalLM loss: logits -> fp32, shift, CE mean.
 # Caller passes ALREADY-SHIFTED logits [T, V] and labels [T].
