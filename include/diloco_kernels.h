@@ -31,13 +31,18 @@ enum dk_dtype { DK_F32 = 0, DK_F16 = 1, DK_BF16 = 2 };
  * forward the reference calls at train_fsdp.py:378 / train_diloco_torch.py:313.
  * y[r,c] = w[c] * (x[r,c] * invrms[r]);  invrms[r] = rsqrt(mean(x[r,:]^2)+eps)
  * x,y,w: dtype; invrms: f32[rows] (saved for bwd). */
-int dk_rmsnorm_fwd(void* y, float* invrms, const void* x, const void* w,
-                   int64_t rows, int64_t cols, float eps, int dtype, dkStream stream);
-/* dx: dtype; dw_partial: f32[grid][cols] workspace written deterministically;
- * grid is returned by dk_rmsnorm_bwd_grid(rows). A second call to
- * dk_reduce_partials sums dw_partial into dw (f32[cols]). */
-int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy, const void* x,
-                   const void* w, const float* invrms,
+/* res/h_out (nullable, paired): fuse the preceding residual add
+ * h = x + res (the reference's `x + self_attn(...)` adds inside the decoder
+ * layer) into the normalisation pass: h is written out, y = norm(h). */
+int dk_rmsnorm_fwd(void* y, void* h_out, float* invrms, const void* x,
+                   const void* res, const void* w, int64_t rows, int64_t cols,
+                   float eps, int dtype, dkStream stream);
+/* dx: dtype; dres (nullable): upstream residual gradient added into dx
+ * (dx == d(residual input) == d(res)); dw_partial: f32[grid][cols] workspace
+ * written deterministically; grid from dk_rmsnorm_bwd_grid(rows). A second
+ * call to dk_reduce_partials sums dw_partial into dw (f32[cols]). */
+int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy, const void* dres,
+                   const void* x, const void* w, const float* invrms,
                    int64_t rows, int64_t cols, int grid, int dtype, dkStream stream);
 int dk_rmsnorm_bwd_grid(int64_t rows);
 /* out[c] = sum_g partial[g][c], fixed order (deterministic), two parallel

@@ -43,13 +43,30 @@ std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, do
   const int64_t rows = x.numel() / cols;
   auto y = at::empty_like(x);
   auto invrms = at::empty({rows}, x.options().dtype(at::kFloat));
-  DK_OK(dk_rmsnorm_fwd(y.data_ptr(), invrms.data_ptr<float>(), x.data_ptr(), w.data_ptr(),
-                       rows, cols, (float)eps, dt_of(x), stream()));
+  DK_OK(dk_rmsnorm_fwd(y.data_ptr(), nullptr, invrms.data_ptr<float>(), x.data_ptr(),
+                       nullptr, w.data_ptr(), rows, cols, (float)eps, dt_of(x), stream()));
   return {y, invrms};
 }
 
-std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
-                                    const at::Tensor& w, const at::Tensor& invrms) {
+std::vector<at::Tensor> rmsnorm_add_fwd(const at::Tensor& x, const at::Tensor& res,
+                                        const at::Tensor& w, double eps) {
+  CHECK_DEV_CONTIG(x);
+  CHECK_DEV_CONTIG(res);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  auto y = at::empty_like(x);
+  auto h = at::empty_like(x);
+  auto invrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  DK_OK(dk_rmsnorm_fwd(y.data_ptr(), h.data_ptr(), invrms.data_ptr<float>(), x.data_ptr(),
+                       res.data_ptr(), w.data_ptr(), rows, cols, (float)eps, dt_of(x),
+                       stream()));
+  return {y, h, invrms};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
+                                    const c10::optional<at::Tensor>& dres,
+                                    const at::Tensor& x, const at::Tensor& w,
+                                    const at::Tensor& invrms) {
   CHECK_DEV_CONTIG(dy);
   CHECK_DEV_CONTIG(x);
   const int64_t cols = x.size(-1);
@@ -59,9 +76,10 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dwp = at::empty({grid, cols}, x.options().dtype(at::kFloat));
   auto tmp = at::empty({dk_reduce_tmp_slices(grid), cols}, x.options().dtype(at::kFloat));
   auto dw = at::empty({cols}, x.options().dtype(at::kFloat));
-  DK_OK(dk_rmsnorm_bwd(dx.data_ptr(), dwp.data_ptr<float>(), dy.data_ptr(), x.data_ptr(),
-                       w.data_ptr(), invrms.data_ptr<float>(), rows, cols, grid,
-                       dt_of(x), stream()));
+  const void* dres_p = dres.has_value() ? dres->data_ptr() : nullptr;
+  DK_OK(dk_rmsnorm_bwd(dx.data_ptr(), dwp.data_ptr<float>(), dy.data_ptr(), dres_p,
+                       x.data_ptr(), w.data_ptr(), invrms.data_ptr<float>(), rows, cols,
+                       grid, dt_of(x), stream()));
   DK_OK(dk_reduce_partials(dw.data_ptr<float>(), tmp.data_ptr<float>(),
                            dwp.data_ptr<float>(), grid, cols, stream()));
   return {dx, dw};
@@ -252,6 +270,7 @@ std::string version() { return dk_version(); }
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_add_fwd", &rmsnorm_add_fwd);
   m.def("rope", &rope);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);

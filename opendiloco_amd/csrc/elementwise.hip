@@ -17,10 +17,14 @@
 // forward: y = w * (x * rsqrt(mean(x^2)+eps)); matches transformers
 // LlamaRMSNorm (fp32 internal, output rounded to input dtype).
 
-template <int DT, int NT>
+// RES: fuse the residual add h = x + res (h written back, rounded to T
+// exactly like a separate torch add would) into the normalisation pass.
+template <int DT, int NT, bool RES>
 __global__ void rmsnorm_fwd_kernel(typename DTraits<DT>::T* __restrict__ y,
+                                   typename DTraits<DT>::T* __restrict__ h_out,
                                    float* __restrict__ invrms,
                                    const typename DTraits<DT>::T* __restrict__ x,
+                                   const typename DTraits<DT>::T* __restrict__ res,
                                    const typename DTraits<DT>::T* __restrict__ w,
                                    int64_t rows, int cols, float eps) {
   using TR = DTraits<DT>;
@@ -31,21 +35,49 @@ __global__ void rmsnorm_fwd_kernel(typename DTraits<DT>::T* __restrict__ y,
 
   const int nvec = cols / W;
   for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
-    const T* xr = x + r * cols;
+    const T* xr = (RES ? h_out : x) + r * cols;  // second pass reads h
     T* yr = y + r * cols;
     float ss = 0.f;
-    for (int i = threadIdx.x; i < nvec; i += NT) {
-      V xv = *(const V*)(xr + i * W);
+    if (RES) {
+      const T* xin = x + r * cols;
+      const T* rin = res + r * cols;
+      T* hr = h_out + r * cols;
+      for (int i = threadIdx.x; i < nvec; i += NT) {
+        V xv = *(const V*)(xin + i * W);
+        V rv = *(const V*)(rin + i * W);
+        V hv;
 #pragma unroll
-      for (int j = 0; j < W; ++j) {
-        float f = TR::toF(((const T*)&xv)[j]);
+        for (int j = 0; j < W; ++j) {
+          T h = TR::fromF(TR::toF(((const T*)&xv)[j]) + TR::toF(((const T*)&rv)[j]));
+          ((T*)&hv)[j] = h;
+          float f = TR::toF(h);
+          ss += f * f;
+        }
+        *(V*)(hr + i * W) = hv;
+      }
+      for (int c = nvec * W + threadIdx.x; c < cols; c += NT) {
+        T h = TR::fromF(TR::toF(xin[c]) + TR::toF(rin[c]));
+        hr[c] = h;
+        float f = TR::toF(h);
+        ss += f * f;
+      }
+    } else {
+      for (int i = threadIdx.x; i < nvec; i += NT) {
+        V xv = *(const V*)(xr + i * W);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float f = TR::toF(((const T*)&xv)[j]);
+          ss += f * f;
+        }
+      }
+      for (int c = nvec * W + threadIdx.x; c < cols; c += NT) {
+        float f = TR::toF(xr[c]);
         ss += f * f;
       }
     }
-    for (int c = nvec * W + threadIdx.x; c < cols; c += NT) {
-      float f = TR::toF(xr[c]);
-      ss += f * f;
-    }
+    if (RES) __syncthreads();  // h_out writes visible to this block's own
+                               // second pass (same thread re-reads its own
+                               // writes; barrier orders sred reuse anyway)
     ss = block_reduce_sum<NT>(ss, sred);
     const float ir = rsqrtf(ss / (float)cols + eps);
     if (threadIdx.x == 0 && invrms) invrms[r] = ir;
@@ -74,10 +106,11 @@ __global__ void rmsnorm_fwd_kernel(typename DTraits<DT>::T* __restrict__ y,
 //   t, t+NT, ...) and are written once at the end — no global RMW, no
 //   zero-init, deterministic (block b owns rows b, b+grid, ...).
 // MAXC = max columns per thread: cols <= NT*MAXC (8192 at NT=256).
-template <int DT, int NT, int MAXC>
+template <int DT, int NT, int MAXC, bool DRES>
 __global__ void rmsnorm_bwd_kernel(typename DTraits<DT>::T* __restrict__ dx,
                                    float* __restrict__ dwp,
                                    const typename DTraits<DT>::T* __restrict__ dy,
+                                   const typename DTraits<DT>::T* __restrict__ dres,
                                    const typename DTraits<DT>::T* __restrict__ x,
                                    const typename DTraits<DT>::T* __restrict__ w,
                                    const float* __restrict__ invrms,
@@ -101,12 +134,15 @@ __global__ void rmsnorm_bwd_kernel(typename DTraits<DT>::T* __restrict__ dx,
       dot += g * xh;
     }
     dot = block_reduce_sum<NT>(dot, sred) / (float)cols;
+    const T* drr = DRES ? dres + r * cols : nullptr;
     int k = 0;
     for (int c = threadIdx.x; c < cols; c += NT, ++k) {
       float xh = TR::toF(xr[c]) * ir;
       float dyf = TR::toF(dyr[c]);
       float g = dyf * TR::toF(w[c]);
-      dxr[c] = TR::fromF((g - xh * dot) * ir);
+      float dxv = (g - xh * dot) * ir;
+      if (DRES) dxv += TR::toF(drr[c]);
+      dxr[c] = TR::fromF(dxv);
       dwacc[k] += dyf * xh;
     }
   }
@@ -147,36 +183,53 @@ __global__ void reduce_partials_stage2(float* __restrict__ out,
 
 extern "C" int dk_rmsnorm_bwd_grid(int64_t rows) { return (int)(rows < 2048 ? rows : 2048); }
 
-extern "C" int dk_rmsnorm_fwd(void* y, float* invrms, const void* x, const void* w,
-                              int64_t rows, int64_t cols, float eps, int dtype,
-                              dkStream stream) {
+extern "C" int dk_rmsnorm_fwd(void* y, void* h_out, float* invrms, const void* x,
+                              const void* res, const void* w, int64_t rows,
+                              int64_t cols, float eps, int dtype, dkStream stream) {
   constexpr int NT = 256;
   int grid = (int)(rows < 4096 ? rows : 4096);
   DK_DISPATCH_DT(dtype, {
     using T = typename DTraits<kDT>::T;
-    hipLaunchKernelGGL((rmsnorm_fwd_kernel<kDT, NT>), dim3(grid), dim3(NT), 0,
-                       (hipStream_t)stream, (T*)y, invrms, (const T*)x, (const T*)w,
-                       rows, (int)cols, eps);
+    if (res != nullptr)
+      hipLaunchKernelGGL((rmsnorm_fwd_kernel<kDT, NT, true>), dim3(grid), dim3(NT), 0,
+                         (hipStream_t)stream, (T*)y, (T*)h_out, invrms, (const T*)x,
+                         (const T*)res, (const T*)w, rows, (int)cols, eps);
+    else
+      hipLaunchKernelGGL((rmsnorm_fwd_kernel<kDT, NT, false>), dim3(grid), dim3(NT), 0,
+                         (hipStream_t)stream, (T*)y, nullptr, invrms, (const T*)x,
+                         nullptr, (const T*)w, rows, (int)cols, eps);
   });
   DK_CHECK_LAUNCH();
   return 0;
 }
 
-extern "C" int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy, const void* x,
-                              const void* w, const float* invrms, int64_t rows,
-                              int64_t cols, int grid, int dtype, dkStream stream) {
+extern "C" int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy,
+                              const void* dres, const void* x, const void* w,
+                              const float* invrms, int64_t rows, int64_t cols,
+                              int grid, int dtype, dkStream stream) {
   constexpr int NT = 256;
   if (cols > NT * 32) return (int)hipErrorInvalidValue;
   DK_DISPATCH_DT(dtype, {
     using T = typename DTraits<kDT>::T;
-    if (cols <= NT * 8)
-      hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 8>), dim3(grid), dim3(NT), 0,
-                         (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
-                         (const T*)x, (const T*)w, invrms, rows, (int)cols);
-    else
-      hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 32>), dim3(grid), dim3(NT), 0,
-                         (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
-                         (const T*)x, (const T*)w, invrms, rows, (int)cols);
+    if (cols <= NT * 8) {
+      if (dres != nullptr)
+        hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 8, true>), dim3(grid), dim3(NT), 0,
+                           (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                           (const T*)dres, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+      else
+        hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 8, false>), dim3(grid), dim3(NT), 0,
+                           (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                           nullptr, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+    } else {
+      if (dres != nullptr)
+        hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 32, true>), dim3(grid), dim3(NT), 0,
+                           (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                           (const T*)dres, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+      else
+        hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 32, false>), dim3(grid), dim3(NT), 0,
+                           (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                           nullptr, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+    }
   });
   DK_CHECK_LAUNCH();
   return 0;

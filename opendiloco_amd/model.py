@@ -246,6 +246,19 @@ class DecoderLayer(nn.Module):
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
+    def forward_fused(self, h: torch.Tensor, delta, rotary: Rotary):
+        """Residual-fused path: takes (h, delta) with the pending residual
+        delta from the previous layer folded into this layer's first norm
+        (one fused kernel instead of add+norm); returns (h', delta')."""
+        eps = self.input_layernorm.variance_epsilon
+        if delta is None:
+            n1 = self.input_layernorm(h)
+        else:
+            n1, h = ops.rmsnorm_add(h, delta, _cast(self.input_layernorm.weight, h.dtype), eps)
+        a = self.self_attn(n1, rotary)
+        n2, h = ops.rmsnorm_add(h, a, _cast(self.post_attention_layernorm.weight, h.dtype), eps)
+        return h, self.mlp(n2)
+
 
 class LlamaBackbone(nn.Module):
     def __init__(self, cfg: LlamaModelConfig):
@@ -335,9 +348,14 @@ class LlamaForCausalLM(nn.Module):
         else:
             h = _CastEmbeddingFn.apply(input_ids, self.model.embed_tokens.weight,
                                        self._cached_embed(cdtype))
+        delta = None
         for layer in self.model.layers:
-            h = layer(h, self.rotary)
-        h = self.model.norm(h)
+            h, delta = layer.forward_fused(h, delta, self.rotary)
+        if delta is None:
+            h = self.model.norm(h)
+        else:
+            h, _ = ops.rmsnorm_add(h, delta, _cast(self.model.norm.weight, cdtype),
+                                   self.model.norm.variance_epsilon)
         logits = self.lm_head(h)  # [B, S, V], compute dtype
 
         loss = None

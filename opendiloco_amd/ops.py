@@ -70,7 +70,7 @@ class _RMSNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, invrms = ctx.saved_tensors
         if x.is_cuda:
-            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), x, w, invrms)
+            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), None, x, w, invrms)
             return dx, dw.to(w.dtype), None
         xf, dyf, wf = x.float(), dy.float(), w.float()
         ir = invrms.reshape(*x.shape[:-1], 1).float()
@@ -84,6 +84,53 @@ class _RMSNormFn(torch.autograd.Function):
 
 def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
     return _RMSNormFn.apply(x, w, eps)
+
+
+class _RMSNormAddFn(torch.autograd.Function):
+    """Fused h = x + res; y = rmsnorm(h).  Returns (y, h); the backward
+    folds the residual-fork gradient (dh from h's downstream use) into dx, so
+    the decoder layer's residual adds cost no separate kernels.  Bitwise
+    equal to the unfused add-then-norm sequence."""
+
+    @staticmethod
+    def forward(ctx, x, res, w, eps):
+        if x.is_cuda:
+            y, h, invrms = _ext().rmsnorm_add_fwd(x.contiguous(), res.contiguous(),
+                                                  w.contiguous(), eps)
+        else:
+            h = x + res
+            hf = h.float()
+            invrms = torch.rsqrt(hf.pow(2).mean(-1) + eps)
+            y = w * (hf * invrms.unsqueeze(-1)).to(h.dtype)
+        ctx.save_for_backward(h, w, invrms)
+        ctx.set_materialize_grads(False)
+        return y, h
+
+    @staticmethod
+    def backward(ctx, dy, dh):
+        h, w, invrms = ctx.saved_tensors
+        if dy is None:  # y unused (cannot happen in the model); fall back
+            dy = torch.zeros_like(h)
+        if h.is_cuda:
+            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), dh.contiguous() if dh is not None else None,
+                                        h, w, invrms)
+            return dx, dx, dw.to(w.dtype), None
+        hf, dyf, wf = h.float(), dy.float(), w.float()
+        ir = invrms.unsqueeze(-1).float()
+        xhat = hf * ir
+        g = dyf * wf
+        dot = (g * xhat).mean(-1, keepdim=True)
+        dx = ((g - xhat * dot) * ir)
+        if dh is not None:
+            dx = dx + dh.float()
+        dx = dx.to(h.dtype)
+        dw = (dyf * xhat).reshape(-1, h.shape[-1]).sum(0).to(w.dtype)
+        return dx, dx, dw, None
+
+
+def rmsnorm_add(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor, eps: float):
+    """(y, h) where h = x + res, y = rmsnorm(h)."""
+    return _RMSNormAddFn.apply(x, res, w, eps)
 
 
 # ====================== RoPE ======================
