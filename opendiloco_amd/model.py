@@ -338,11 +338,15 @@ class LlamaForCausalLM(nn.Module):
                 labels: torch.Tensor | None = None, **_ignored) -> CausalLMOutput:
         device = input_ids.device
         cdtype = self._dtype_for(device)
-        if attention_mask is not None and not torch.all(attention_mask.bool()):
-            raise NotImplementedError(
-                "padding attention masks are not supported: the hot path is the "
-                "reference's fake/C4 full-sequence batches (mask all ones, "
-                "open_diloco/utils.py:166)")
+        # checked once per model (the .all() forces a device sync; the hot
+        # loop must stay async)
+        if attention_mask is not None and not getattr(self, "_mask_checked", False):
+            self._mask_checked = True
+            if not bool(attention_mask.bool().all()):
+                raise NotImplementedError(
+                    "padding attention masks are not supported: the hot path is "
+                    "the reference's fake/C4 full-sequence batches (mask all "
+                    "ones, open_diloco/utils.py:166)")
         if self.rotary.cos.device != device:
             self.rotary.to_(device)
         S = input_ids.shape[1]
