@@ -78,16 +78,18 @@ int dk_swiglu2_bwd(void* dgu, const void* dy, const void* gu, int64_t rows,
 
 /* ---- Fused cross-entropy ------------------------------------------------
  * Replaces the causal-LM loss of LlamaForCausalLM (logits->fp32, shifted CE,
- * mean over tokens) invoked via model(**batch) at train_fsdp.py:378.
- * logits: dtype[T, V]; labels: int64[T]; loss_rows/lse: f32[T]. */
+ * mean over tokens) invoked via model(**batch) at train_fsdp.py:378.  The
+ * causal shift is internal: logits dtype[B, S, V], labels int64[B, S];
+ * position s < S-1 scores against labels[b, s+1]; loss_rows/lse are
+ * f32[B*(S-1)]. */
 int dk_cross_entropy_fwd(float* loss_rows, float* lse, const void* logits,
-                         const int64_t* labels, int64_t T, int64_t V,
+                         const int64_t* labels, int64_t B, int64_t S, int64_t V,
                          int dtype, dkStream stream);
-/* dlogits = (softmax(logits) - onehot(labels)) * (*dloss);  dloss: f32 device
- * scalar (upstream grad / T folded in by caller or via scale_per_row). */
+/* dlogits (full [B, S, V], last position zeroed) =
+ * (softmax(logits) - onehot(shifted labels)) * (*dloss) * inv_T. */
 int dk_cross_entropy_bwd(void* dlogits, const void* logits, const float* lse,
                          const int64_t* labels, const float* dloss, float inv_T,
-                         int64_t T, int64_t V, int dtype, dkStream stream);
+                         int64_t B, int64_t S, int64_t V, int dtype, dkStream stream);
 
 /* ---- Flash attention (causal, GQA) --------------------------------------
  * Replaces torch SDPA inside the reference's model forward (attn_implementation

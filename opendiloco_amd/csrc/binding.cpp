@@ -137,12 +137,13 @@ at::Tensor swiglu2_bwd(const at::Tensor& dy, const at::Tensor& gu, int64_t I) {
 // ---- cross entropy ----
 std::vector<at::Tensor> ce_fwd(const at::Tensor& logits, const at::Tensor& labels) {
   CHECK_DEV_CONTIG(logits);
-  const int64_t V = logits.size(-1);
-  const int64_t T = logits.numel() / V;
+  TORCH_CHECK(logits.dim() == 3, "ce_fwd expects [B, S, V] logits");
+  const int64_t B = logits.size(0), S = logits.size(1), V = logits.size(2);
+  const int64_t T = B * (S - 1);
   auto loss_rows = at::empty({T}, logits.options().dtype(at::kFloat));
   auto lse = at::empty({T}, logits.options().dtype(at::kFloat));
   DK_OK(dk_cross_entropy_fwd(loss_rows.data_ptr<float>(), lse.data_ptr<float>(),
-                             logits.data_ptr(), labels.data_ptr<int64_t>(), T, V,
+                             logits.data_ptr(), labels.data_ptr<int64_t>(), B, S, V,
                              dt_of(logits), stream()));
   return {loss_rows, lse};
 }
@@ -150,11 +151,10 @@ std::vector<at::Tensor> ce_fwd(const at::Tensor& logits, const at::Tensor& label
 at::Tensor ce_bwd(const at::Tensor& logits, const at::Tensor& lse, const at::Tensor& labels,
                   const at::Tensor& dloss, double inv_T) {
   auto dlogits = at::empty_like(logits);
-  const int64_t V = logits.size(-1);
-  const int64_t T = logits.numel() / V;
+  const int64_t B = logits.size(0), S = logits.size(1), V = logits.size(2);
   DK_OK(dk_cross_entropy_bwd(dlogits.data_ptr(), logits.data_ptr(), lse.data_ptr<float>(),
                              labels.data_ptr<int64_t>(), dloss.data_ptr<float>(),
-                             (float)inv_T, T, V, dt_of(logits), stream()));
+                             (float)inv_T, B, S, V, dt_of(logits), stream()));
   return dlogits;
 }
 

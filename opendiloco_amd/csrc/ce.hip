@@ -3,9 +3,12 @@
 // Replaces the loss computation inside transformers LlamaForCausalLM
 // (logits -> fp32, shifted CE with mean reduction) that the reference hits
 // via model(**batch) at train_fsdp.py:378 / train_diloco_torch.py:313.
-// The caller passes ALREADY-SHIFTED rows: logits [T, V] vs labels [T] where
-// T = B*(S-1).  Forward is a single online-softmax pass (one HBM read of the
-// 64 KB row); backward one read + one write.  No atomics — deterministic.
+// The causal SHIFT lives inside the kernels: the caller passes the FULL
+// [B, S, V] logits and [B, S] labels; row (b, s<S-1) scores logits[b,s,:]
+// against labels[b,s+1] (T = B*(S-1) rows in the mean), and the backward
+// writes the full [B, S, V] gradient with the last position zeroed — no
+// 2.1 GB slice copy / pad round trips.  Forward is a single online-softmax
+// pass; backward one read + one write.  No atomics — deterministic.
 
 #include "dk_common.h"
 #include "../../include/diloco_kernels.h"
@@ -23,7 +26,8 @@ __device__ __forceinline__ void merge_ms(float& m, float& s, float om, float os)
 template <int DT, int NT>
 __global__ void ce_fwd_kernel(float* __restrict__ loss_rows, float* __restrict__ lse_out,
                               const typename DTraits<DT>::T* __restrict__ logits,
-                              const int64_t* __restrict__ labels, int64_t T, int64_t V) {
+                              const int64_t* __restrict__ labels, int64_t B, int64_t Sm1,
+                              int64_t S, int64_t V) {
   using TR = DTraits<DT>;
   using TT = typename TR::T;
   constexpr int W = VecIO<TT>::W;
@@ -31,8 +35,11 @@ __global__ void ce_fwd_kernel(float* __restrict__ loss_rows, float* __restrict__
   __shared__ float sm[NT / DK_WAVE], ss[NT / DK_WAVE];
 
   const int64_t nvec = V / W;
+  const int64_t T = B * Sm1;
   for (int64_t r = blockIdx.x; r < T; r += gridDim.x) {
-    const TT* row = logits + r * V;
+    const int64_t b = r / Sm1, sp = r % Sm1;   // scores pos sp vs label sp+1
+    const TT* row = logits + (b * S + sp) * V;
+    const int64_t lab_r = labels[b * S + sp + 1];
     float m = -INFINITY, s = 0.f;
     for (int64_t i = threadIdx.x; i < nvec; i += NT) {
       V8 xv = *(const V8*)(row + i * W);
@@ -63,7 +70,7 @@ __global__ void ce_fwd_kernel(float* __restrict__ loss_rows, float* __restrict__
       for (int i = 1; i < NT / DK_WAVE; ++i) merge_ms(m, s, sm[i], ss[i]);
       float lse = m + __logf(s);
       lse_out[r] = lse;
-      loss_rows[r] = lse - TR::toF(row[labels[r]]);
+      loss_rows[r] = lse - TR::toF(row[lab_r]);
     }
     __syncthreads();
   }
@@ -75,18 +82,30 @@ __global__ void ce_bwd_kernel(typename DTraits<DT>::T* __restrict__ dlogits,
                               const float* __restrict__ lse,
                               const int64_t* __restrict__ labels,
                               const float* __restrict__ dloss, float inv_T,
-                              int64_t T, int64_t V) {
+                              int64_t B, int64_t Sm1, int64_t S, int64_t V) {
   using TR = DTraits<DT>;
   using TT = typename TR::T;
   constexpr int W = VecIO<TT>::W;
   using V8 = typename VecIO<TT>::V;
   const float scale = dloss[0] * inv_T;
   const int64_t nvec = V / W;
-  for (int64_t r = blockIdx.x; r < T; r += gridDim.x) {
-    const TT* row = logits + r * V;
-    TT* drow = dlogits + r * V;
+  const int64_t rows_all = B * S;  // covers the zeroed last position too
+  for (int64_t rr = blockIdx.x; rr < rows_all; rr += gridDim.x) {
+    const int64_t b = rr / S, sp = rr % S;
+    TT* drow0 = dlogits + rr * V;
+    if (sp == S - 1) {  // no next-token target: zero gradient row
+      for (int64_t i = threadIdx.x; i < nvec; i += NT) {
+        V8 z = (V8)(0);
+        *(V8*)(drow0 + i * W) = z;
+      }
+      for (int64_t c = nvec * W + threadIdx.x; c < V; c += NT) drow0[c] = TR::fromF(0.f);
+      continue;
+    }
+    const int64_t r = b * Sm1 + sp;
+    const TT* row = logits + rr * V;
+    TT* drow = drow0;
     const float l = lse[r];
-    const int64_t lab = labels[r];
+    const int64_t lab = labels[rr + 1];
     for (int64_t i = threadIdx.x; i < nvec; i += NT) {
       V8 xv = *(const V8*)(row + i * W);
       V8 dv;
@@ -107,14 +126,16 @@ __global__ void ce_bwd_kernel(typename DTraits<DT>::T* __restrict__ dlogits,
 }
 
 extern "C" int dk_cross_entropy_fwd(float* loss_rows, float* lse, const void* logits,
-                                    const int64_t* labels, int64_t T, int64_t V,
-                                    int dtype, dkStream stream) {
+                                    const int64_t* labels, int64_t B, int64_t S,
+                                    int64_t V, int dtype, dkStream stream) {
   constexpr int NT = 256;
+  const int64_t T = B * (S - 1);
   int grid = (int)(T < 2048 ? T : 2048);
   DK_DISPATCH_DT(dtype, {
     using TT = typename DTraits<kDT>::T;
     hipLaunchKernelGGL((ce_fwd_kernel<kDT, NT>), dim3(grid), dim3(NT), 0,
-                       (hipStream_t)stream, loss_rows, lse, (const TT*)logits, labels, T, V);
+                       (hipStream_t)stream, loss_rows, lse, (const TT*)logits, labels,
+                       B, S - 1, S, V);
   });
   DK_CHECK_LAUNCH();
   return 0;
@@ -122,15 +143,16 @@ extern "C" int dk_cross_entropy_fwd(float* loss_rows, float* lse, const void* lo
 
 extern "C" int dk_cross_entropy_bwd(void* dlogits, const void* logits, const float* lse,
                                     const int64_t* labels, const float* dloss,
-                                    float inv_T, int64_t T, int64_t V, int dtype,
-                                    dkStream stream) {
+                                    float inv_T, int64_t B, int64_t S, int64_t V,
+                                    int dtype, dkStream stream) {
   constexpr int NT = 256;
-  int grid = (int)(T < 2048 ? T : 2048);
+  const int64_t rows_all = B * S;
+  int grid = (int)(rows_all < 2048 ? rows_all : 2048);
   DK_DISPATCH_DT(dtype, {
     using TT = typename DTraits<kDT>::T;
     hipLaunchKernelGGL((ce_bwd_kernel<kDT, NT>), dim3(grid), dim3(NT), 0,
                        (hipStream_t)stream, (TT*)dlogits, (const TT*)logits, lse, labels,
-                       dloss, inv_T, T, V);
+                       dloss, inv_T, B, S - 1, S, V);
   });
   DK_CHECK_LAUNCH();
   return 0;

@@ -369,11 +369,9 @@ class LlamaForCausalLM(nn.Module):
 
         loss = None
         if labels is not None:
-            # HF shift: predict token t+1 from logits at t; mean over B*(S-1)
-            V = logits.shape[-1]
-            shift_logits = logits[:, :-1, :].reshape(-1, V)
-            shift_labels = labels[:, 1:].reshape(-1)
-            loss = ops.causal_lm_loss(shift_logits.contiguous(), shift_labels.contiguous())
+            # HF shift (predict token t+1 from logits at t; mean over
+            # B*(S-1)) happens inside the fused CE kernel
+            loss = ops.causal_lm_loss(logits, labels)
         return CausalLMOutput(loss=loss, logits=logits)
 
     def train_config_summary(self) -> dict:

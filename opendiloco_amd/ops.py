@@ -235,21 +235,24 @@ def swiglu_fused(gu: torch.Tensor, inter: int) -> torch.Tensor:
 
 # ====================== causal-LM cross entropy ======================
 # transformers LlamaForCausalLM loss: logits -> fp32, shift, CE mean.
-# Caller passes ALREADY-SHIFTED logits [T, V] and labels [T].
+# Takes the FULL [B, S, V] logits and [B, S] labels; the causal shift
+# (position s scores labels[s+1], T = B*(S-1)) happens inside the kernel —
+# no 2.1 GB slice copy / grad pad round trips.
 
 class _CrossEntropyFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels):
-        V = logits.shape[-1]
-        T = logits.numel() // V
+        B, S, V = logits.shape
+        T = B * (S - 1)
         if logits.is_cuda:
             loss_rows, lse = _ext().ce_fwd(logits.contiguous(), labels.contiguous())
             ctx.save_for_backward(logits, lse, labels)
             ctx.T = T
             return loss_rows.mean()
-        lf = logits.float()
+        lf = logits[:, :-1, :].float().reshape(T, V)
+        lab = labels[:, 1:].reshape(T)
         lse = torch.logsumexp(lf, dim=-1)
-        loss_rows = lse - lf.gather(-1, labels.unsqueeze(-1)).squeeze(-1)
+        loss_rows = lse - lf.gather(-1, lab.unsqueeze(-1)).squeeze(-1)
         ctx.save_for_backward(logits, lse, labels)
         ctx.T = T
         return loss_rows.mean()
@@ -261,13 +264,20 @@ class _CrossEntropyFn(torch.autograd.Function):
         if logits.is_cuda:
             dlogits = _ext().ce_bwd(logits, lse, labels, dloss.contiguous().float(), 1.0 / T)
             return dlogits, None
-        p = torch.softmax(logits.float(), dim=-1)
-        p.scatter_add_(-1, labels.unsqueeze(-1), torch.full_like(labels, -1, dtype=p.dtype).unsqueeze(-1))
-        return (p * (dloss.float() / T)).to(logits.dtype), None
+        B, S, V = logits.shape
+        lf = logits[:, :-1, :].float().reshape(T, V)
+        lab = labels[:, 1:].reshape(T)
+        p = torch.softmax(lf, dim=-1)
+        p.scatter_add_(-1, lab.unsqueeze(-1),
+                       torch.full_like(lab, -1, dtype=p.dtype).unsqueeze(-1))
+        d = (p * (dloss.float() / T)).to(logits.dtype).reshape(B, S - 1, V)
+        dlogits = torch.zeros_like(logits)
+        dlogits[:, :-1, :] = d
+        return dlogits, None
 
 
 def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-    """Mean CE over pre-shifted rows (logits [T,V] vs labels [T])."""
+    """Shifted mean CE over full [B, S, V] logits vs [B, S] labels."""
     return _CrossEntropyFn.apply(logits, labels)
 
 

@@ -126,21 +126,25 @@ def test_swiglu_fwd_bwd(ext):
 # ---------------- cross entropy ----------------
 
 @requires_gpu
-@pytest.mark.parametrize("T,V", [(256, 1024), (100, 32000), (7, 1000)])
-def test_cross_entropy_fwd_bwd(ext, T, V):
+@pytest.mark.parametrize("B,S,V", [(4, 65, 1024), (2, 51, 32000), (1, 8, 1000)])
+def test_cross_entropy_fwd_bwd(ext, B, S, V):
+    """Fused CE with internal causal shift: position s scores labels[s+1];
+    grad of the last position is zero."""
     from opendiloco_amd import ops
 
-    logits = torch.randn(T, V) * 3
-    labels = torch.randint(0, V, (T,))
+    logits = torch.randn(B, S, V) * 3
+    labels = torch.randint(0, V, (B, S))
     lg = logits.to("cuda", torch.bfloat16).contiguous().requires_grad_(True)
     lab = labels.cuda()
     loss = ops.causal_lm_loss(lg, lab)
     lf = lg.detach().float().requires_grad_(True)
-    lref = torch.nn.functional.cross_entropy(lf, lab)
+    lref = torch.nn.functional.cross_entropy(
+        lf[:, :-1, :].reshape(-1, V), lab[:, 1:].reshape(-1))
     assert loss.item() == pytest.approx(lref.item(), rel=1e-3, abs=1e-3)
     loss.backward()
     lref.backward()
     assert torch.allclose(lg.grad.float(), lf.grad, atol=1e-4, rtol=1e-2)
+    assert lg.grad[:, -1, :].abs().max().item() == 0.0
 
 
 # ---------------- attention ----------------
