@@ -61,6 +61,15 @@ int dk_reduce_partials(float* out, float* tmp, const float* partial, int grid,
 int dk_rope(void* out, const void* x, const float* costab, const float* sintab,
             int64_t n_rows, int64_t S, int64_t D, int backward, int dtype,
             dkStream stream);
+/* generic-stride RoPE gather/scatter: moves a head-tensor between the fused
+ * QKV buffer layout (addr = b*sb + h*sh + s*sr) and contiguous [B,H,S,D],
+ * applying the rotation (rotate=1; backward=1 -> transposed) or plain copy
+ * (rotate=0, the V path) in the same pass. */
+int dk_rope_move(void* out, const void* in, const float* costab, const float* sintab,
+                 int64_t B, int64_t H, int64_t S, int64_t D,
+                 int64_t i_sb, int64_t i_sh, int64_t i_sr,
+                 int64_t o_sb, int64_t o_sh, int64_t o_sr,
+                 int backward, int rotate, int dtype, dkStream stream);
 
 /* ---- SwiGLU -------------------------------------------------------------
  * Replaces transformers LlamaMLP's act_fn(gate)*up (silu).
@@ -95,20 +104,30 @@ int dk_cross_entropy_bwd(void* dlogits, const void* logits, const float* lse,
  * Replaces torch SDPA inside the reference's model forward (attn_implementation
  * "sdpa", train_fsdp.py:107).  Layout [B, H, S, D] contiguous; D in {32,64};
  * o: dtype; lse: f32[B,Hq,S] = m + log(l) saved for bwd.  scale = 1/sqrt(D). */
+/* o_sb/o_sh/o_sr (and g_* on the backward entries): stride triplet
+ * (addr = b*sb + h*sh + s*sr) for the attention output / upstream gradient,
+ * so o can be written directly in [B, S, Hq*D] and do read from it without
+ * transpose copies; sb == 0 selects the contiguous [B,H,S,D] default. */
 int dk_attn_fwd(void* o, float* lse, const void* q, const void* k, const void* v,
                 int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
-                float scale, int dtype, dkStream stream);
+                float scale, int64_t o_sb, int64_t o_sh, int64_t o_sr,
+                int dtype, dkStream stream);
 /* delta[b,h,s] = rowsum(do * o), fp32 — preprocessing for bwd. */
 int dk_attn_bwd_preprocess(float* delta, const void* do_, const void* o,
-                           int64_t rows, int64_t D, int dtype, dkStream stream);
+                           int64_t B, int64_t H, int64_t S, int64_t D,
+                           int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                           int dtype, dkStream stream);
 int dk_attn_bwd_dkdv(void* dk_out, void* dv_out, const void* do_, const void* q,
                      const void* k, const void* v, const float* lse,
                      const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
-                     int64_t S, int64_t D, float scale, int dtype, dkStream stream);
+                     int64_t S, int64_t D, float scale,
+                     int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                     int dtype, dkStream stream);
 int dk_attn_bwd_dq(void* dq_out, const void* do_, const void* q, const void* k,
                    const void* v, const float* lse, const float* delta,
                    int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
-                   float scale, int dtype, dkStream stream);
+                   float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                   int dtype, dkStream stream);
 
 /* ---- Fused AdamW (inner optimizer) --------------------------------------
  * Replaces the torch AdamW step the reference runs at hivemind_diloco.py:546-550

@@ -89,7 +89,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const typename DTraits<DT>::T* __restrict__ q,
     const typename DTraits<DT>::T* __restrict__ k,
     const typename DTraits<DT>::T* __restrict__ v,
-    int B, int Hq, int Hkv, int S, float scale) {
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t o_sb, int64_t o_sh, int64_t o_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA16<DT>;
@@ -266,9 +267,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const int qrow = q0 + hi * 4 + r;
     if (qrow >= S) continue;
     const float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+    const int64_t obase = (int64_t)b * o_sb + (int64_t)h * o_sh + (int64_t)qrow * o_sr;
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn)
-      o[qoff + (int64_t)qrow * D + dn * 16 + lo] = TR::fromF(o_acc[dn][r] * inv_l);
+      o[obase + dn * 16 + lo] = TR::fromF(o_acc[dn][r] * inv_l);
     if (lo == 0) lse[((int64_t)b * Hq + h) * S + qrow] = m_run[r] + __logf(l_run[r]);
   }
 }
@@ -278,16 +280,21 @@ template <int DT>
 __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
                                     const typename DTraits<DT>::T* __restrict__ do_,
                                     const typename DTraits<DT>::T* __restrict__ o,
-                                    int64_t rows, int D) {
+                                    int64_t rows, int H, int S, int D,
+                                    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
   using TR = DTraits<DT>;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int64_t w0 = (int64_t)blockIdx.x * 4 + wave;
   const int64_t stride = (int64_t)gridDim.x * 4;
-  for (int64_t r = w0; r < rows; r += stride) {
+  for (int64_t r = w0; r < rows; r += stride) {  // r in (b, h, s) order
+    const int sp = (int)(r % S);
+    const int hh = (int)((r / S) % H);
+    const int64_t b = r / ((int64_t)S * H);
+    const int64_t base = b * g_sb + hh * g_sh + sp * g_sr;
     float s = 0.f;
     for (int c = lane; c < D; c += 64)
-      s += TR::toF(do_[r * D + c]) * TR::toF(o[r * D + c]);
+      s += TR::toF(do_[base + c]) * TR::toF(o[base + c]);
     s = wave_reduce_sum(s);
     if (lane == 0) delta[r] = s;
   }
@@ -308,7 +315,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     const typename DTraits<DT>::T* __restrict__ k,
     const typename DTraits<DT>::T* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    int B, int Hq, int Hkv, int S, float scale) {
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA16<DT>;
@@ -376,12 +384,13 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   frag qreg, dreg;
   float lse_reg = 0.f, dl_reg = 0.f;
 
+  const int64_t gbase = (int64_t)b * g_sb + (int64_t)h * g_sh;
   auto load_qtile = [&](int qt) {
     const int qrow = qt * QT + st_row;
     const int qr_c = qrow < S ? qrow : S - 1;
     if (st_on) {
       qreg = *(const frag*)(q + qoff + (int64_t)qr_c * D + st_c8);
-      dreg = *(const frag*)(do_ + qoff + (int64_t)qr_c * D + st_c8);
+      dreg = *(const frag*)(do_ + gbase + (int64_t)qr_c * g_sr + st_c8);
     }
     if (st_t < QT) {
       const int rr = qt * QT + st_t;
@@ -510,7 +519,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const typename DTraits<DT>::T* __restrict__ k,
     const typename DTraits<DT>::T* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    int B, int Hq, int Hkv, int S, float scale) {
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA16<DT>;
@@ -549,10 +559,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   {
     const int qrow = q0 + lo;
     const int qr_c = qrow < S ? qrow : S - 1;
+    const int64_t gbase = (int64_t)b * g_sb + (int64_t)h * g_sh;
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       q_frag[kc] = *(const frag*)(q + qoff + (int64_t)qr_c * D + kc * 32 + hi * 8);
-      do_frag[kc] = *(const frag*)(do_ + qoff + (int64_t)qr_c * D + kc * 32 + hi * 8);
+      do_frag[kc] = *(const frag*)(do_ + gbase + (int64_t)qr_c * g_sr + kc * 32 + hi * 8);
     }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -664,7 +675,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 template <int DT, int D>
 static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
                            const void* v, int64_t B, int64_t Hq, int64_t Hkv,
-                           int64_t S, float scale, dkStream stream) {
+                           int64_t S, float scale, int64_t o_sb, int64_t o_sh,
+                           int64_t o_sr, dkStream stream) {
   using T = typename DTraits<DT>::T;
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
@@ -672,29 +684,35 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
   const size_t lds = sizeof(T) * (2 * KT * DS + 2 * D * KS + 4 * 16 * KS);
   hipLaunchKernelGGL((attn_fwd_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
-                     (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale);
+                     (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale,
+                     o_sb, o_sh, o_sr);
   DK_CHECK_LAUNCH();
   return 0;
 }
 
 extern "C" int dk_attn_fwd(void* o, float* lse, const void* q, const void* k,
                            const void* v, int64_t B, int64_t Hq, int64_t Hkv,
-                           int64_t S, int64_t D, float scale, int dtype,
+                           int64_t S, int64_t D, float scale,
+                           int64_t o_sb, int64_t o_sh, int64_t o_sr, int dtype,
                            dkStream stream) {
   if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
+  if (o_sb == 0) { o_sb = Hq * S * D; o_sh = S * D; o_sr = D; }  // BHSD default
   if (D == 64) {
-    if (dtype == 2) return launch_attn_fwd<2, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
-    return launch_attn_fwd<1, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
+    if (dtype == 2) return launch_attn_fwd<2, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
+    return launch_attn_fwd<1, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
   } else if (D == 32) {
-    if (dtype == 2) return launch_attn_fwd<2, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
-    return launch_attn_fwd<1, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, stream);
+    if (dtype == 2) return launch_attn_fwd<2, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
+    return launch_attn_fwd<1, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
   }
   return (int)hipErrorInvalidValue;
 }
 
 extern "C" int dk_attn_bwd_preprocess(float* delta, const void* do_, const void* o,
-                                      int64_t rows, int64_t D, int dtype,
-                                      dkStream stream) {
+                                      int64_t B, int64_t H, int64_t S, int64_t D,
+                                      int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                                      int dtype, dkStream stream) {
+  const int64_t rows = B * H * S;
+  if (g_sb == 0) { g_sb = H * S * D; g_sh = S * D; g_sr = D; }
   int64_t g = (rows + 3) / 4;  // 4 waves per block, one row per wave-iteration
   int grid = (int)(g > 2048 ? 2048 : (g < 1 ? 1 : g));
   DK_DISPATCH_DT(dtype, {
@@ -702,7 +720,7 @@ extern "C" int dk_attn_bwd_preprocess(float* delta, const void* do_, const void*
       using T = typename DTraits<kDT>::T;
       hipLaunchKernelGGL((attn_bwd_pre_kernel<kDT>), dim3(grid), dim3(256), 0,
                          (hipStream_t)stream, delta, (const T*)do_, (const T*)o,
-                         rows, (int)D);
+                         rows, (int)H, (int)S, (int)D, g_sb, g_sh, g_sr);
     } else {
       return (int)hipErrorInvalidValue;
     }
@@ -715,7 +733,8 @@ template <int DT, int D>
 static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const void* q,
                                 const void* k, const void* v, const float* lse,
                                 const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
-                                int64_t S, float scale, dkStream stream) {
+                                int64_t S, float scale, int64_t g_sb, int64_t g_sh,
+                                int64_t g_sr, dkStream stream) {
   using T = typename DTraits<DT>::T;
   const int nKT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nKT);
@@ -724,7 +743,7 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
   hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
                      (const T*)q, (const T*)k, (const T*)v, lse, delta,
-                     (int)B, (int)Hq, (int)Hkv, (int)S, scale);
+                     (int)B, (int)Hq, (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
   DK_CHECK_LAUNCH();
   return 0;
 }
@@ -732,15 +751,17 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
 extern "C" int dk_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const void* q,
                                 const void* k, const void* v, const float* lse,
                                 const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
-                                int64_t S, int64_t D, float scale, int dtype,
+                                int64_t S, int64_t D, float scale,
+                                int64_t g_sb, int64_t g_sh, int64_t g_sr, int dtype,
                                 dkStream stream) {
   if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
+  if (g_sb == 0) { g_sb = Hq * S * D; g_sh = S * D; g_sr = D; }
   if (D == 64) {
-    if (dtype == 2) return launch_attn_bwd_dkdv<2, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
-    return launch_attn_bwd_dkdv<1, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    if (dtype == 2) return launch_attn_bwd_dkdv<2, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    return launch_attn_bwd_dkdv<1, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
   } else if (D == 32) {
-    if (dtype == 2) return launch_attn_bwd_dkdv<2, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
-    return launch_attn_bwd_dkdv<1, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    if (dtype == 2) return launch_attn_bwd_dkdv<2, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    return launch_attn_bwd_dkdv<1, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
   }
   return (int)hipErrorInvalidValue;
 }
@@ -749,7 +770,8 @@ template <int DT, int D>
 static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const void* k,
                               const void* v, const float* lse, const float* delta,
                               int64_t B, int64_t Hq, int64_t Hkv, int64_t S,
-                              float scale, dkStream stream) {
+                              float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                              dkStream stream) {
   using T = typename DTraits<DT>::T;
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
@@ -758,7 +780,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
-                     (int)Hkv, (int)S, scale);
+                     (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
   DK_CHECK_LAUNCH();
   return 0;
 }
@@ -766,14 +788,16 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
 extern "C" int dk_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const void* k,
                               const void* v, const float* lse, const float* delta,
                               int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
-                              float scale, int dtype, dkStream stream) {
+                              float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                              int dtype, dkStream stream) {
   if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
+  if (g_sb == 0) { g_sb = Hq * S * D; g_sh = S * D; g_sr = D; }
   if (D == 64) {
-    if (dtype == 2) return launch_attn_bwd_dq<2, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
-    return launch_attn_bwd_dq<1, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    if (dtype == 2) return launch_attn_bwd_dq<2, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    return launch_attn_bwd_dq<1, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
   } else if (D == 32) {
-    if (dtype == 2) return launch_attn_bwd_dq<2, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
-    return launch_attn_bwd_dq<1, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, stream);
+    if (dtype == 2) return launch_attn_bwd_dq<2, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    return launch_attn_bwd_dq<1, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
   }
   return (int)hipErrorInvalidValue;
 }

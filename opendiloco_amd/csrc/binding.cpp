@@ -169,7 +169,8 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   auto o = at::empty_like(q);
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   DK_OK(dk_attn_fwd(o.data_ptr(), lse.data_ptr<float>(), q.data_ptr(), k.data_ptr(),
-                    v.data_ptr(), B, Hq, Hkv, S, D, (float)scale, dt_of(q), stream()));
+                    v.data_ptr(), B, Hq, Hkv, S, D, (float)scale, 0, 0, 0,
+                    dt_of(q), stream()));
   return {o, lse};
 }
 
@@ -181,7 +182,7 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& do_, const at::Tensor& q,
   const int64_t Hkv = k.size(1);
   auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   DK_OK(dk_attn_bwd_preprocess(delta.data_ptr<float>(), do_.data_ptr(), o.data_ptr(),
-                               B * Hq * S, D, dt_of(q), stream()));
+                               B, Hq, S, D, 0, 0, 0, dt_of(q), stream()));
   auto dq = at::empty_like(q);
   // dk/dv computed per q-head; python caller sums GQA groups when Hq != Hkv
   auto dk_full = at::empty({B, Hq, S, D}, q.options());
@@ -189,10 +190,76 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& do_, const at::Tensor& q,
   DK_OK(dk_attn_bwd_dkdv(dk_full.data_ptr(), dv_full.data_ptr(), do_.data_ptr(),
                          q.data_ptr(), k.data_ptr(), v.data_ptr(), lse.data_ptr<float>(),
                          delta.data_ptr<float>(), B, Hq, Hkv, S, D, (float)scale,
-                         dt_of(q), stream()));
+                         0, 0, 0, dt_of(q), stream()));
   DK_OK(dk_attn_bwd_dq(dq.data_ptr(), do_.data_ptr(), q.data_ptr(), k.data_ptr(),
                        v.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       B, Hq, Hkv, S, D, (float)scale, dt_of(q), stream()));
+                       B, Hq, Hkv, S, D, (float)scale, 0, 0, 0, dt_of(q), stream()));
+  return {dq, dk_full, dv_full};
+}
+
+// ---- fused QKV attention path: o in [B, S, Hq*D]; do read back the same way;
+// rope gather/scatter between the packed qkv buffer and BHSD head tensors ----
+at::Tensor qkv_rope_gather(const at::Tensor& qkv, const at::Tensor& costab,
+                           const at::Tensor& sintab, int64_t H, int64_t D,
+                           int64_t col_off, bool rotate) {
+  CHECK_DEV_CONTIG(qkv);
+  const int64_t B = qkv.size(0), S = qkv.size(1), TOT = qkv.size(2);
+  auto out = at::empty({B, H, S, D}, qkv.options());
+  DK_OK(dk_rope_move(out.data_ptr(), (const char*)qkv.data_ptr() + col_off * qkv.element_size(),
+                     costab.data_ptr<float>(), sintab.data_ptr<float>(), B, H, S, D,
+                     /*in*/ S * TOT, D, TOT, /*out*/ H * S * D, S * D, D,
+                     0, rotate ? 1 : 0, dt_of(qkv), stream()));
+  return out;
+}
+
+void rope_scatter_(at::Tensor& dqkv, const at::Tensor& src_bhsd, const at::Tensor& costab,
+                   const at::Tensor& sintab, int64_t col_off, bool rotate) {
+  CHECK_DEV_CONTIG(dqkv);
+  CHECK_DEV_CONTIG(src_bhsd);
+  const int64_t B = dqkv.size(0), S = dqkv.size(1), TOT = dqkv.size(2);
+  const int64_t H = src_bhsd.size(1), D = src_bhsd.size(3);
+  DK_OK(dk_rope_move((char*)dqkv.data_ptr() + col_off * dqkv.element_size(),
+                     src_bhsd.data_ptr(),
+                     costab.data_ptr<float>(), sintab.data_ptr<float>(), B, H, S, D,
+                     /*in*/ H * S * D, S * D, D, /*out*/ S * TOT, D, TOT,
+                     1, rotate ? 1 : 0, dt_of(dqkv), stream()));
+}
+
+std::vector<at::Tensor> attn_fwd_bsd(const at::Tensor& q, const at::Tensor& k,
+                                     const at::Tensor& v, double scale) {
+  CHECK_DEV_CONTIG(q);
+  const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
+  const int64_t Hkv = k.size(1);
+  auto o = at::empty({B, S, Hq * D}, q.options());
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  DK_OK(dk_attn_fwd(o.data_ptr(), lse.data_ptr<float>(), q.data_ptr(), k.data_ptr(),
+                    v.data_ptr(), B, Hq, Hkv, S, D, (float)scale,
+                    /*o strides (b,h,s)*/ S * Hq * D, D, Hq * D, dt_of(q), stream()));
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd_bsd(const at::Tensor& do_bsd, const at::Tensor& q,
+                                     const at::Tensor& k, const at::Tensor& v,
+                                     const at::Tensor& o_bsd, const at::Tensor& lse,
+                                     double scale) {
+  CHECK_DEV_CONTIG(do_bsd);
+  const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
+  const int64_t Hkv = k.size(1);
+  const int64_t sb = S * Hq * D, sh = D, sr = Hq * D;
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  DK_OK(dk_attn_bwd_preprocess(delta.data_ptr<float>(), do_bsd.data_ptr(),
+                               o_bsd.data_ptr(), B, Hq, S, D, sb, sh, sr,
+                               dt_of(q), stream()));
+  auto dq = at::empty_like(q);
+  auto dk_full = at::empty({B, Hq, S, D}, q.options());
+  auto dv_full = at::empty({B, Hq, S, D}, q.options());
+  DK_OK(dk_attn_bwd_dkdv(dk_full.data_ptr(), dv_full.data_ptr(), do_bsd.data_ptr(),
+                         q.data_ptr(), k.data_ptr(), v.data_ptr(), lse.data_ptr<float>(),
+                         delta.data_ptr<float>(), B, Hq, Hkv, S, D, (float)scale,
+                         sb, sh, sr, dt_of(q), stream()));
+  DK_OK(dk_attn_bwd_dq(dq.data_ptr(), do_bsd.data_ptr(), q.data_ptr(), k.data_ptr(),
+                       v.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       B, Hq, Hkv, S, D, (float)scale, sb, sh, sr, dt_of(q), stream()));
   return {dq, dk_full, dv_full};
 }
 
@@ -280,6 +347,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("qkv_rope_gather", &qkv_rope_gather);
+  m.def("rope_scatter_", &rope_scatter_);
+  m.def("attn_fwd_bsd", &attn_fwd_bsd);
+  m.def("attn_bwd_bsd", &attn_bwd_bsd);
   m.def("fused_adamw", &fused_adamw);
   m.def("clip_grad_", &clip_grad_);
   m.def("grad_norm", &grad_norm);

@@ -295,6 +295,89 @@ __global__ void rope_kernel(typename DTraits<DT>::T* __restrict__ out,
   }
 }
 
+// rope_move: generic-stride RoPE gather/scatter between the fused QKV
+// buffer layout (addr = b*sb + h*sh + s*sr + col) and contiguous [B,H,S,D].
+// ROT=0 copies without rotation (the V path); BWD applies the transposed
+// rotation (gradient path).  Replaces the per-layer transpose+contiguous
+// copies AND the separate rope pass in one kernel.
+template <int DT, int BWD, int ROT>
+__global__ void rope_move_kernel(typename DTraits<DT>::T* __restrict__ out,
+                                 const typename DTraits<DT>::T* __restrict__ in,
+                                 const float* __restrict__ costab,
+                                 const float* __restrict__ sintab,
+                                 int64_t total_rows, int H, int S, int D,
+                                 int64_t i_sb, int64_t i_sh, int64_t i_sr,
+                                 int64_t o_sb, int64_t o_sh, int64_t o_sr) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  const int hd = D / 2;
+  const int nq = hd / 4;  // 4-pair quads per row
+  const int64_t total = total_rows * nq;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = idx / nq;
+    const int i0 = (int)(idx % nq) * 4;
+    const int sp = (int)(row % S);
+    const int hh = (int)((row / S) % H);
+    const int64_t b = row / ((int64_t)S * H);
+    const T* irow = in + b * i_sb + hh * i_sh + sp * i_sr;
+    T* orow = out + b * o_sb + hh * o_sh + sp * o_sr;
+    shortx4 x1 = *(const shortx4*)(irow + i0);
+    shortx4 x2 = *(const shortx4*)(irow + i0 + hd);
+    if (ROT) {
+      const float4 cv = *(const float4*)(costab + (int64_t)sp * hd + i0);
+      const float4 sv = *(const float4*)(sintab + (int64_t)sp * hd + i0);
+      shortx4 o1, o2;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float a = TR::toF(((const T*)&x1)[j]);
+        float bb = TR::toF(((const T*)&x2)[j]);
+        float c = ((const float*)&cv)[j];
+        float sn = BWD ? -((const float*)&sv)[j] : ((const float*)&sv)[j];
+        ((T*)&o1)[j] = TR::fromF(a * c - bb * sn);
+        ((T*)&o2)[j] = TR::fromF(bb * c + a * sn);
+      }
+      *(shortx4*)(orow + i0) = o1;
+      *(shortx4*)(orow + i0 + hd) = o2;
+    } else {
+      *(shortx4*)(orow + i0) = x1;
+      *(shortx4*)(orow + i0 + hd) = x2;
+    }
+  }
+}
+
+extern "C" int dk_rope_move(void* out, const void* in, const float* costab,
+                            const float* sintab, int64_t B, int64_t H, int64_t S,
+                            int64_t D, int64_t i_sb, int64_t i_sh, int64_t i_sr,
+                            int64_t o_sb, int64_t o_sh, int64_t o_sr, int backward,
+                            int rotate, int dtype, dkStream stream) {
+  if ((D / 2) % 4 != 0 || dtype == 0) return (int)hipErrorInvalidValue;
+  const int64_t total_rows = B * H * S;
+  int grid = dk_stream_grid(total_rows * (D / 2) / 4, 256);
+  DK_DISPATCH_DT(dtype, {
+    if constexpr (kDT != 0) {
+      using T = typename DTraits<kDT>::T;
+      if (rotate && backward)
+        hipLaunchKernelGGL((rope_move_kernel<kDT, 1, 1>), dim3(grid), dim3(256), 0,
+                           (hipStream_t)stream, (T*)out, (const T*)in, costab, sintab,
+                           total_rows, (int)H, (int)S, (int)D, i_sb, i_sh, i_sr,
+                           o_sb, o_sh, o_sr);
+      else if (rotate)
+        hipLaunchKernelGGL((rope_move_kernel<kDT, 0, 1>), dim3(grid), dim3(256), 0,
+                           (hipStream_t)stream, (T*)out, (const T*)in, costab, sintab,
+                           total_rows, (int)H, (int)S, (int)D, i_sb, i_sh, i_sr,
+                           o_sb, o_sh, o_sr);
+      else
+        hipLaunchKernelGGL((rope_move_kernel<kDT, 0, 0>), dim3(grid), dim3(256), 0,
+                           (hipStream_t)stream, (T*)out, (const T*)in, costab, sintab,
+                           total_rows, (int)H, (int)S, (int)D, i_sb, i_sh, i_sr,
+                           o_sb, o_sh, o_sr);
+    }
+  });
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
 // f32 variant uses scalar pairs (rare path, CPU-parity testing only)
 template <int BWD>
 __global__ void rope_kernel_f32(float* __restrict__ out, const float* __restrict__ x,

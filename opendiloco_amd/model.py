@@ -205,6 +205,11 @@ class Attention(nn.Module):
         B, S, _ = x.shape
         D = self.head_dim
         qkv = self._qkv(x)
+        if x.is_cuda:
+            # fused: rope-gather out of the packed buffer, o written [B,S,Hq*D]
+            o = ops.qkv_rope_attention(qkv, rotary.cos, rotary.sin,
+                                       self.num_heads, self.num_kv_heads, D, self.scale)
+            return self.o_proj(o)
         nq, nkv = self.num_heads * D, self.num_kv_heads * D
         q, k, v = qkv.split([nq, nkv, nkv], dim=-1)
         q = q.view(B, S, self.num_heads, D).transpose(1, 2).contiguous()

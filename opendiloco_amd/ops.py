@@ -353,3 +353,48 @@ def _attn_cpu_bwd(do, q, k, v, o, lse, scale):
 
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float) -> torch.Tensor:
     return _AttentionFn.apply(q, k, v, scale)
+
+
+class _QKVRopeAttentionFn(torch.autograd.Function):
+    """GPU-only fused path over the packed QKV projection output
+    [B, S, (Hq+2*Hkv)*D]: RoPE-gather q/k (+ plain gather v) straight out of
+    the packed buffer into [B,H,S,D], flash attention writing o directly as
+    [B, S, Hq*D], and a backward that RoPE-scatters dq/dk/dv back into ONE
+    dqkv buffer — no transpose+contiguous copies, no split/cat."""
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, Hq, Hkv, D, scale):
+        ext = _ext()
+        nq, nkv = Hq * D, Hkv * D
+        q = ext.qkv_rope_gather(qkv, cos, sin, Hq, D, 0, True)
+        k = ext.qkv_rope_gather(qkv, cos, sin, Hkv, D, nq, True)
+        v = ext.qkv_rope_gather(qkv, cos, sin, Hkv, D, nq + nkv, False)
+        o_bsd, lse = ext.attn_fwd_bsd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o_bsd, lse, cos, sin)
+        ctx.dims = (Hq, Hkv, D, scale)
+        return o_bsd
+
+    @staticmethod
+    def backward(ctx, do_bsd):
+        q, k, v, o_bsd, lse, cos, sin = ctx.saved_tensors
+        Hq, Hkv, D, scale = ctx.dims
+        ext = _ext()
+        dq, dkf, dvf = ext.attn_bwd_bsd(do_bsd.contiguous(), q, k, v, o_bsd, lse, scale)
+        if Hq != Hkv:
+            g = Hq // Hkv
+            B, _, S, _ = q.shape
+            dk = dkf.view(B, Hkv, g, S, D).sum(2).contiguous()
+            dv = dvf.view(B, Hkv, g, S, D).sum(2).contiguous()
+        else:
+            dk, dv = dkf, dvf
+        B, _, S, _ = q.shape
+        nq, nkv = Hq * D, Hkv * D
+        dqkv = torch.empty(B, S, nq + 2 * nkv, dtype=q.dtype, device=q.device)
+        ext.rope_scatter_(dqkv, dq, cos, sin, 0, True)
+        ext.rope_scatter_(dqkv, dk, cos, sin, nq, True)
+        ext.rope_scatter_(dqkv, dv, cos, sin, nq + nkv, False)
+        return dqkv, None, None, None, None, None, None
+
+
+def qkv_rope_attention(qkv, cos, sin, Hq, Hkv, D, scale):
+    return _QKVRopeAttentionFn.apply(qkv, cos, sin, Hq, Hkv, D, scale)
