@@ -276,6 +276,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 }
 
 // ======================= bwd preprocess: delta = rowsum(do*o) =======================
+// each wave covers WR=4 rows per iteration: 16 lanes per row, shortx4 loads
+// (8 B/lane), 4-lane... 16-lane-group shuffle reduce per row
 template <int DT>
 __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
                                     const typename DTraits<DT>::T* __restrict__ do_,
@@ -283,20 +285,32 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
                                     int64_t rows, int H, int S, int D,
                                     int64_t g_sb, int64_t g_sh, int64_t g_sr) {
   using TR = DTraits<DT>;
+  using T = typename TR::T;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
+  const int lo = lane & 15, hi = lane >> 4;   // lo: D-quad index, hi: row-in-group
   const int64_t w0 = (int64_t)blockIdx.x * 4 + wave;
   const int64_t stride = (int64_t)gridDim.x * 4;
-  for (int64_t r = w0; r < rows; r += stride) {  // r in (b, h, s) order
-    const int sp = (int)(r % S);
-    const int hh = (int)((r / S) % H);
-    const int64_t b = r / ((int64_t)S * H);
+  const int quads = D / 4;  // shortx4 chunks per row (D in {32, 64})
+  const int64_t groups = (rows + 3) / 4;
+  for (int64_t g = w0; g < groups; g += stride) {  // 4 rows per wave-iteration
+    const int64_t r = g * 4 + hi;
+    const int64_t rc = r < rows ? r : rows - 1;
+    const int sp = (int)(rc % S);
+    const int hh = (int)((rc / S) % H);
+    const int64_t b = rc / ((int64_t)S * H);
     const int64_t base = b * g_sb + hh * g_sh + sp * g_sr;
     float s = 0.f;
-    for (int c = lane; c < D; c += 64)
-      s += TR::toF(do_[base + c]) * TR::toF(o[base + c]);
-    s = wave_reduce_sum(s);
-    if (lane == 0) delta[r] = s;
+    for (int q4 = lo; q4 < quads; q4 += 16) {
+      shortx4 dv = *(const shortx4*)(do_ + base + q4 * 4);
+      shortx4 ov = *(const shortx4*)(o + base + q4 * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        s += TR::toF(((const T*)&dv)[j]) * TR::toF(((const T*)&ov)[j]);
+    }
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, DK_WAVE);
+    if (lo == 0 && r < rows) delta[r] = s;
   }
 }
 

@@ -181,6 +181,65 @@ __global__ void reduce_partials_stage2(float* __restrict__ out,
   }
 }
 
+// vectorized rmsnorm backward for cols == NT*VEC (the 150m/1b hidden sizes):
+// thread t owns the contiguous columns [t*VEC, (t+1)*VEC); w is loaded into
+// registers ONCE per block; each row is read once (values kept in registers
+// between the dot pass and the dx/dw pass) with shortx4/8 vector loads.
+template <int DT, int NT, int VEC, bool DRES>
+__global__ void rmsnorm_bwd_vec_kernel(typename DTraits<DT>::T* __restrict__ dx,
+                                       float* __restrict__ dwp,
+                                       const typename DTraits<DT>::T* __restrict__ dy,
+                                       const typename DTraits<DT>::T* __restrict__ dres,
+                                       const typename DTraits<DT>::T* __restrict__ x,
+                                       const typename DTraits<DT>::T* __restrict__ w,
+                                       const float* __restrict__ invrms,
+                                       int64_t rows, int cols) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  typedef __attribute__((ext_vector_type(VEC))) short vec_t;
+  __shared__ float sred[NT / DK_WAVE];
+  const int c0 = threadIdx.x * VEC;
+  float wv[VEC], dwacc[VEC];
+  {
+    vec_t wvv = *(const vec_t*)(w + c0);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) { wv[j] = TR::toF(((const T*)&wvv)[j]); dwacc[j] = 0.f; }
+  }
+  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+    const float ir = invrms[r];
+    float xf[VEC], dyf[VEC];
+    {
+      vec_t xv = *(const vec_t*)(x + r * cols + c0);
+      vec_t dv = *(const vec_t*)(dy + r * cols + c0);
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        xf[j] = TR::toF(((const T*)&xv)[j]);
+        dyf[j] = TR::toF(((const T*)&dv)[j]);
+      }
+    }
+    float s1 = 0.f;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) s1 += dyf[j] * wv[j] * xf[j];
+    const float dot = block_reduce_sum<NT>(s1, sred) * ir / (float)cols;
+    vec_t dxv;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float xh = xf[j] * ir;
+      float g = dyf[j] * wv[j];
+      float dv = (g - xh * dot) * ir;
+      if (DRES) dv += TR::toF(dres[r * cols + c0 + j]);
+      ((T*)&dxv)[j] = TR::fromF(dv);
+      dwacc[j] += dyf[j] * xh;
+    }
+    *(vec_t*)(dx + r * cols + c0) = dxv;
+  }
+  {
+    float* dwrow = dwp + (int64_t)blockIdx.x * cols;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) dwrow[c0 + j] = dwacc[j];
+  }
+}
+
 extern "C" int dk_rmsnorm_bwd_grid(int64_t rows) { return (int)(rows < 2048 ? rows : 2048); }
 
 extern "C" int dk_rmsnorm_fwd(void* y, void* h_out, float* invrms, const void* x,
@@ -211,6 +270,33 @@ extern "C" int dk_rmsnorm_bwd(void* dx, float* dw_partial, const void* dy,
   if (cols > NT * 32) return (int)hipErrorInvalidValue;
   DK_DISPATCH_DT(dtype, {
     using T = typename DTraits<kDT>::T;
+    if constexpr (kDT != 0) {
+      // vectorized path for the hot hidden sizes
+      if (cols == NT * 4) {
+        if (dres != nullptr)
+          hipLaunchKernelGGL((rmsnorm_bwd_vec_kernel<kDT, NT, 4, true>), dim3(grid), dim3(NT), 0,
+                             (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                             (const T*)dres, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+        else
+          hipLaunchKernelGGL((rmsnorm_bwd_vec_kernel<kDT, NT, 4, false>), dim3(grid), dim3(NT), 0,
+                             (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                             nullptr, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+        hipError_t e = hipGetLastError();
+        return (int)e;
+      }
+      if (cols == NT * 8) {
+        if (dres != nullptr)
+          hipLaunchKernelGGL((rmsnorm_bwd_vec_kernel<kDT, NT, 8, true>), dim3(grid), dim3(NT), 0,
+                             (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                             (const T*)dres, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+        else
+          hipLaunchKernelGGL((rmsnorm_bwd_vec_kernel<kDT, NT, 8, false>), dim3(grid), dim3(NT), 0,
+                             (hipStream_t)stream, (T*)dx, dw_partial, (const T*)dy,
+                             nullptr, (const T*)x, (const T*)w, invrms, rows, (int)cols);
+        hipError_t e = hipGetLastError();
+        return (int)e;
+      }
+    }
     if (cols <= NT * 8) {
       if (dres != nullptr)
         hipLaunchKernelGGL((rmsnorm_bwd_kernel<kDT, NT, 8, true>), dim3(grid), dim3(NT), 0,
