@@ -70,7 +70,7 @@ def bench_rmsnorm(ext, reps):
     print(f"rmsnorm_fwd [{rows}x{cols}]: {dt*1e3:.3f} ms, {algo/dt/1e9:.0f} GB/s")
     y = torch.randn_like(x)
     ir = torch.rand(rows, device="cuda") + 0.5
-    dt = _timeit(lambda: ext.rmsnorm_bwd(y, x, w, ir), reps)
+    dt = _timeit(lambda: ext.rmsnorm_bwd(y, None, x, w, ir), reps)
     print(f"rmsnorm_bwd: {dt*1e3:.3f} ms, {3*rows*cols*2/dt/1e9:.0f} GB/s (3x bf16 tensors)")
 
 
