@@ -307,3 +307,113 @@ def test_cast(ext):
     dst = torch.empty(12345, device="cuda", dtype=torch.bfloat16)
     ext.cast_(dst, src)
     assert torch.allclose(dst.float(), src, atol=0.02, rtol=0.01)
+
+
+# ---------------- edge cases / properties ----------------
+
+@requires_gpu
+def test_attention_minimum_shapes(ext):
+    """Smallest tiles: S=64, B=1, single head."""
+    from opendiloco_amd import ops
+
+    q, k, v = _to_dev_bf16(torch.randn(1, 1, 64, 64), torch.randn(1, 1, 64, 64),
+                           torch.randn(1, 1, 64, 64))
+    for t in (q, k, v):
+        t.requires_grad_(True)
+    o = ops.attention(q, k, v, 0.125)
+    qf = q.detach().float()
+    s = (qf @ k.detach().float().transpose(-1, -2)) * 0.125
+    s = s.masked_fill(torch.triu(torch.ones(64, 64, dtype=torch.bool, device="cuda"), 1),
+                      float("-inf"))
+    oref = torch.softmax(s, -1) @ v.detach().float()
+    assert torch.allclose(o.float(), oref, atol=3e-2, rtol=3e-2)
+    o.backward(torch.ones_like(o))
+    assert torch.isfinite(q.grad.float()).all()
+
+
+@requires_gpu
+def test_attention_linearity_in_v(ext):
+    """Attention output is linear in V at fixed Q,K (softmax weights fixed):
+    attn(q,k, a*v1 + b*v2) == a*attn(q,k,v1) + b*attn(q,k,v2)."""
+    from opendiloco_amd import ops
+
+    torch.manual_seed(5)
+    q, k = _to_dev_bf16(torch.randn(1, 2, 128, 64), torch.randn(1, 2, 128, 64))
+    v1, v2 = _to_dev_bf16(torch.randn(1, 2, 128, 64), torch.randn(1, 2, 128, 64))
+    o1 = ops.attention(q, k, v1, 0.125).float()
+    o2 = ops.attention(q, k, v2, 0.125).float()
+    v3 = (2.0 * v1.float() - 0.5 * v2.float()).to(torch.bfloat16)
+    o3 = ops.attention(q, k, v3, 0.125).float()
+    assert torch.allclose(o3, 2.0 * o1 - 0.5 * o2, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+def test_qkv_fused_path_matches_composed(ext, fixture_2m):
+    """The fused qkv_rope_attention equals the composed split+rope+attention
+    pipeline on the same packed buffer (fwd and bwd)."""
+    from opendiloco_amd import ops
+
+    torch.manual_seed(6)
+    B, S, Hq, Hkv, D = 2, 128, 4, 2, 64
+    nq, nkv = Hq * D, Hkv * D
+    qkv = torch.randn(B, S, nq + 2 * nkv, device="cuda", dtype=torch.bfloat16)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2, dtype=torch.float32) / D))
+    f = torch.outer(torch.arange(S, dtype=torch.float32), inv)
+    cos, sin = f.cos().cuda(), f.sin().cuda()
+
+    a = qkv.clone().requires_grad_(True)
+    o_fused = ops.qkv_rope_attention(a, cos, sin, Hq, Hkv, D, D ** -0.5)
+
+    b = qkv.clone().requires_grad_(True)
+    q, k, v = b.split([nq, nkv, nkv], dim=-1)
+    q = q.view(B, S, Hq, D).transpose(1, 2).contiguous()
+    k = k.view(B, S, Hkv, D).transpose(1, 2).contiguous()
+    v = v.view(B, S, Hkv, D).transpose(1, 2).contiguous()
+    q = ops.rope(q, cos, sin, S)
+    k = ops.rope(k, cos, sin, S)
+    o_ref = ops.attention(q, k, v, D ** -0.5).transpose(1, 2).reshape(B, S, nq)
+    assert torch.allclose(o_fused.float(), o_ref.float(), atol=2e-2, rtol=2e-2)
+
+    g = torch.randn_like(o_fused)
+    o_fused.backward(g)
+    o_ref.backward(g)
+    md = (a.grad.float() - b.grad.float()).abs().max().item()
+    assert md < 5e-2, md
+
+
+@requires_gpu
+def test_ce_vocab_not_multiple_of_8(ext):
+    from opendiloco_amd import ops
+
+    logits = torch.randn(2, 17, 1003, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    labels = torch.randint(0, 1003, (2, 17), device="cuda")
+    loss = ops.causal_lm_loss(logits, labels)
+    lf = logits.detach().float().requires_grad_(True)
+    lref = torch.nn.functional.cross_entropy(lf[:, :-1, :].reshape(-1, 1003),
+                                             labels[:, 1:].reshape(-1))
+    assert loss.item() == pytest.approx(lref.item(), rel=1e-3, abs=1e-3)
+    loss.backward()
+    lref.backward()
+    assert torch.allclose(logits.grad.float(), lf.grad, atol=1e-4, rtol=1e-2)
+
+
+@requires_gpu
+def test_rmsnorm_add_fused_matches_unfused(ext):
+    """rmsnorm_add == add-then-rmsnorm bitwise (the fusion claim)."""
+    from opendiloco_amd import ops
+
+    torch.manual_seed(7)
+    x, res = _to_dev_bf16(torch.randn(256, 1024), torch.randn(256, 1024))
+    w = (torch.randn(1024).abs() + 0.5).to("cuda", torch.bfloat16)
+    x1, r1 = x.clone().requires_grad_(True), res.clone().requires_grad_(True)
+    y1, h1 = ops.rmsnorm_add(x1, r1, w, 1e-5)
+    x2, r2 = x.clone().requires_grad_(True), res.clone().requires_grad_(True)
+    h2 = x2 + r2
+    y2 = ops.rmsnorm(h2, w, 1e-5)
+    assert torch.equal(h1, h2.detach())
+    assert torch.equal(y1, y2.detach())
+    g1, g2 = torch.randn_like(y1), torch.randn_like(y1)
+    (y1 * g1 + h1 * g2).sum().backward()
+    (y2 * g1 + h2 * g2).sum().backward()
+    assert torch.allclose(x1.grad.float(), x2.grad.float(), atol=1e-2, rtol=1e-2)
+    assert torch.equal(x1.grad, r1.grad)

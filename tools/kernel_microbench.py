@@ -94,15 +94,16 @@ def bench_rope(ext, reps):
 
 
 def bench_ce(ext, reps):
-    T, V = 8 * 1023, 32000
-    logits = torch.randn(T, V, device="cuda", dtype=torch.bfloat16)
-    labels = torch.randint(0, V, (T,), device="cuda")
+    B, S, V = 8, 1024, 32000
+    T = B * (S - 1)
+    logits = torch.randn(B, S, V, device="cuda", dtype=torch.bfloat16)
+    labels = torch.randint(0, V, (B, S), device="cuda")
     dt = _timeit(lambda: ext.ce_fwd(logits, labels), reps)
-    print(f"ce_fwd [{T}x{V}]: {dt*1e3:.3f} ms, {T*V*2/dt/1e9:.0f} GB/s (1 read)")
+    print(f"ce_fwd [{B}x{S}x{V}]: {dt*1e3:.3f} ms, {T*V*2/dt/1e9:.0f} GB/s (1 read)")
     lse = torch.randn(T, device="cuda")
     dl = torch.ones((), device="cuda")
     dt = _timeit(lambda: ext.ce_bwd(logits, lse, labels, dl, 1.0 / T), reps)
-    print(f"ce_bwd: {dt*1e3:.3f} ms, {2*T*V*2/dt/1e9:.0f} GB/s (r+w)")
+    print(f"ce_bwd: {dt*1e3:.3f} ms, {2*B*S*V*2/dt/1e9:.0f} GB/s (r+w)")
 
 
 def bench_attn(ext, reps):
