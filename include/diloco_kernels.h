@@ -176,6 +176,9 @@ int dk_probe_mfma_16x16x32_bf16(float* out_d, const void* a16x32, const void* b3
                                 dkStream stream);
 int dk_probe_mfma_16x16x32_bf16_alt(float* out_d, const void* a16x32, const void* b32x16,
                                     dkStream stream);
+int dk_probe_mfma_32x32x16_bf16(float* out_d, const void* a32x16, const void* b16x32,
+                                dkStream stream);
+int dk_probe_permlane32(int* out_d, dkStream stream);
 
 /* version / build info */
 const char* dk_version(void);

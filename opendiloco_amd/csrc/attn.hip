@@ -872,3 +872,50 @@ extern "C" int dk_probe_mfma_16x16x32_bf16_alt(float* out_d, const void* a16x32,
   DK_CHECK_LAUNCH();
   return 0;
 }
+
+// 32x32x16 bf16 probe — assumed maps:
+//   A[32][16]: lane holds row = l&31, k = 8*(l>>5) + j (j = 0..7)
+//   B[16][32]: lane holds col = l&31, k = 8*(l>>5) + j
+//   C[32][32]: lane holds col = l&31, row = (r&3) + 8*(r>>2) + 4*(l>>5)
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+__global__ void probe_mfma32_kernel(float* __restrict__ out,
+                                    const unsigned short* __restrict__ a,
+                                    const unsigned short* __restrict__ b) {
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 31, hi = lane >> 5;
+  shortx8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ((unsigned short*)&af)[j] = a[lo * 16 + hi * 8 + j];
+    ((unsigned short*)&bf)[j] = b[(hi * 8 + j) * 32 + lo];
+  }
+  floatx16 c = (floatx16)(0.f);
+  c = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    out[((r & 3) + 8 * (r >> 2) + 4 * hi) * 32 + lo] = c[r];
+}
+
+extern "C" int dk_probe_mfma_32x32x16_bf16(float* out_d, const void* a32x16,
+                                           const void* b16x32, dkStream stream) {
+  hipLaunchKernelGGL(probe_mfma32_kernel, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                     out_d, (const unsigned short*)a32x16, (const unsigned short*)b16x32);
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
+// permlane32_swap semantics probe: in a[lane] = lane, b[lane] = 100+lane;
+// writes the two results so the host can read the exact lane exchange.
+__global__ void probe_permlane_kernel(int* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  int a = lane, b = 100 + lane;
+  auto r = __builtin_amdgcn_permlane32_swap(a, b, false, false);
+  out[lane] = r[0];
+  out[64 + lane] = r[1];
+}
+
+extern "C" int dk_probe_permlane32(int* out_d, dkStream stream) {
+  hipLaunchKernelGGL(probe_permlane_kernel, dim3(1), dim3(64), 0, (hipStream_t)stream, out_d);
+  DK_CHECK_LAUNCH();
+  return 0;
+}

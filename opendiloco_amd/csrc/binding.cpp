@@ -330,6 +330,20 @@ at::Tensor probe_mfma_alt(const at::Tensor& a, const at::Tensor& b) {
   return out;
 }
 
+at::Tensor probe_mfma32(const at::Tensor& a, const at::Tensor& b) {
+  TORCH_CHECK(a.numel() == 32 * 16 && b.numel() == 16 * 32);
+  auto out = at::zeros({32, 32}, a.options().dtype(at::kFloat));
+  DK_OK(dk_probe_mfma_32x32x16_bf16(out.data_ptr<float>(), a.data_ptr(), b.data_ptr(),
+                                    stream()));
+  return out;
+}
+
+at::Tensor probe_permlane32(const at::Tensor& ref_dev) {
+  auto out = at::zeros({2, 64}, ref_dev.options().dtype(at::kInt));
+  DK_OK(dk_probe_permlane32(out.data_ptr<int>(), stream()));
+  return out;
+}
+
 std::string version() { return dk_version(); }
 
 }  // namespace
@@ -359,5 +373,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cast_", &cast_);
   m.def("probe_mfma", &probe_mfma);
   m.def("probe_mfma_alt", &probe_mfma_alt);
+  m.def("probe_mfma32", &probe_mfma32);
+  m.def("probe_permlane32", &probe_permlane32);
   m.def("version", &version);
 }
