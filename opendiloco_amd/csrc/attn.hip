@@ -30,6 +30,7 @@
 #include "../../include/diloco_kernels.h"
 
 #include <math.h>
+#include <stdlib.h>
 
 typedef __attribute__((ext_vector_type(8))) _Float16 halfx8;
 
@@ -272,6 +273,234 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     for (int dn = 0; dn < NDN; ++dn)
       o[obase + dn * 16 + lo] = TR::fromF(o_acc[dn][r] * inv_l);
     if (lo == 0) lse[((int64_t)b * Hq + h) * S + qrow] = m_run[r] + __logf(l_run[r]);
+  }
+}
+
+// ======================= forward v3: swapped QK^T, in-register softmax =======================
+// 32x32x16 MFMA tiles.  S^T = mfma(K, Q) puts 16 of a q-row's 32 key-scores
+// in ONE lane (q = lane&31; the other 16 in the partner lane l^32), so the
+// softmax row reduction is an in-lane tree + ONE shfl_xor(32) — no 16-lane
+// shuffle chains, no P LDS round trip: P is packed to bf16 in-register
+// (cvt_pk pairs) and redistributed with permlane32_swap into the PV B-frag
+// (layouts + swap semantics hardware-verified by dk_probe_mfma_32x32x16 and
+// dk_probe_permlane32).  PV computes O^T = mfma(V^T, P) so the accumulator
+// stays q-lane-local (alpha rescale in-lane).  Wave owns 32 q rows; LDS is
+// only the double-buffered K/V^T tiles (same async staging as v2).
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+typedef __attribute__((ext_vector_type(4))) int intx4;
+
+template <int DT> struct MFMA32;
+template <> struct MFMA32<2> {
+  using frag = shortx8;
+  static __device__ __forceinline__ floatx16 mma(frag a, frag b, floatx16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <> struct MFMA32<1> {
+  using frag = halfx8;
+  static __device__ __forceinline__ floatx16 mma(frag a, frag b, floatx16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16(a, b, c, 0, 0, 0);
+  }
+};
+
+template <int DT, int D>
+__global__ __launch_bounds__(256) void attn_fwd_v3_kernel(
+    typename DTraits<DT>::T* __restrict__ o, float* __restrict__ lse,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t o_sb, int64_t o_sh, int64_t o_sr) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA32<DT>;
+  using frag = typename MF::frag;
+  constexpr int KT = 64;            // kv tile (2 x 32-key subtiles)
+  constexpr int KS = KT + 8;        // V^T image stride (u16), XOR-swizzled
+  constexpr int DS = D + 8;         // K image stride
+  constexpr int NKC = D / 16;       // 16-channel contraction chunks for QK^T
+  constexpr int NMT = D / 32;       // 32-row d tiles for O^T
+  constexpr int LPT = (KT * D) / 8 / 256;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* K_lds = (T*)smem_raw;                       // [2][KT][DS]
+  T* VT_lds = K_lds + 2 * KT * DS;               // [2][D][KS] swizzled
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo32 = lane & 31;       // q row within the wave
+  const int hi5 = lane >> 5;
+
+  const int nQT = (S + 127) / 128;
+  int bid = blockIdx.x;
+  const int qt = bid % nQT;
+  const int h = (bid / nQT) % Hq;
+  const int b = bid / (nQT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int q0 = qt * 128 + wave * 32;           // wave's first q row
+  const int qrow = q0 + lo32;                    // this lane's q row
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+
+  // persistent Q B-fragments: slot j of chunk kc = Q[qrow][kc*16 + hi5*8 + j]
+  frag q_frag[NKC];
+  {
+    const int qr_c = qrow < S ? qrow : S - 1;
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc)
+      q_frag[kc] = *(const frag*)(q + qoff + (int64_t)qr_c * D + kc * 16 + hi5 * 8);
+  }
+
+  float m_run = NEG_BIG, l_run = 0.f;
+  floatx16 oacc[NMT];
+#pragma unroll
+  for (int mt = 0; mt < NMT; ++mt) oacc[mt] = (floatx16)(0.f);
+
+  const int kv_end = min(S, qt * 128 + 128);
+  const int n_kt = (kv_end + KT - 1) / KT;
+
+  // async double-buffered staging (same structure as v2)
+  shortx8 kreg[LPT], vreg[LPT];
+  const int st_row[2] = {(int)threadIdx.x / (D / 8), (int)(threadIdx.x + 256) / (D / 8)};
+  const int st_c8[2] = {((int)threadIdx.x % (D / 8)) * 8, ((int)(threadIdx.x + 256) % (D / 8)) * 8};
+
+  auto load_tile = [&](int kt) {
+#pragma unroll
+    for (int i = 0; i < LPT; ++i) {
+      const int krow = kt * KT + st_row[i];
+      const int kr_c = krow < S ? krow : S - 1;
+      kreg[i] = *(const shortx8*)(k + kvoff + (int64_t)kr_c * D + st_c8[i]);
+      vreg[i] = *(const shortx8*)(v + kvoff + (int64_t)kr_c * D + st_c8[i]);
+    }
+  };
+  auto write_tile = [&](int buf) {
+    T* Kb = K_lds + buf * KT * DS;
+    T* Vb = VT_lds + buf * D * KS;
+#pragma unroll
+    for (int i = 0; i < LPT; ++i) {
+      *(shortx8*)(Kb + st_row[i] * DS + st_c8[i]) = kreg[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *vt_addr(Vb, st_c8[i] + j, st_row[i] * 2, KS * 2) = ((const T*)&vreg[i])[j];
+    }
+  };
+
+  load_tile(0);
+  write_tile(0);
+  __syncthreads();
+
+  for (int kt = 0; kt < n_kt; ++kt) {
+    const int kbase = kt * KT;
+    const int cur = kt & 1;
+    T* Kb = K_lds + cur * KT * DS;
+    T* Vb = VT_lds + cur * D * KS;
+    if (kt + 1 < n_kt) load_tile(kt + 1);
+
+#pragma unroll
+    for (int st = 0; st < 2; ++st) {            // two 32-key subtiles
+      const int sbase = kbase + st * 32;
+      // ---- S^T = K Q^T ----
+      floatx16 sc = (floatx16)(0.f);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kc = 0; kc < NKC; ++kc) {
+        frag ka = *(const frag*)(Kb + (st * 32 + lo32) * DS + kc * 16 + hi5 * 8);
+        sc = MF::mma(ka, q_frag[kc], sc);
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- in-lane online softmax for q row `qrow` ----
+      const bool full_tile = (sbase + 32 <= q0) && (sbase + 32 <= S);
+      float p[16];
+      float rm = NEG_BIG;
+      if (full_tile) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          p[r] = sc[r] * scale;
+          rm = fmaxf(rm, p[r]);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int key = sbase + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+          p[r] = (key > qrow || key >= S) ? NEG_BIG : sc[r] * scale;
+          rm = fmaxf(rm, p[r]);
+        }
+      }
+      rm = fmaxf(rm, __shfl_xor(rm, 32, DK_WAVE));
+      const float m_new = fmaxf(m_run, rm);
+      const float alpha = __expf(m_run - m_new);
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[r] = p[r] <= NEG_BIG ? 0.f : __expf(p[r] - m_new);
+        psum += p[r];
+      }
+      psum += __shfl_xor(psum, 32, DK_WAVE);
+      l_run = l_run * alpha + psum;
+      m_run = m_new;
+#pragma unroll
+      for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[mt][r] *= alpha;
+
+      // ---- pack P into PV B-fragments via cvt_pk + permlane32_swap ----
+      // chunk c covers keys sbase + c*16 .. +15; per chunk the four packed
+      // words are w0=(k0,k1) w1=(k2,k3) w2=(k8,k9) w3=(k10,k11) in the low
+      // half (+4 in the high half); swap(w0,w2) -> frag dwords {d0, d2},
+      // swap(w1,w3) -> {d1, d3}  (hardware-verified regrouping).
+      frag pfrag[2];
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        const int base = c * 8;
+        unsigned w0 = (unsigned)(unsigned short)TR::fromF(p[base + 0])
+                      | ((unsigned)(unsigned short)TR::fromF(p[base + 1]) << 16);
+        unsigned w1 = (unsigned)(unsigned short)TR::fromF(p[base + 2])
+                      | ((unsigned)(unsigned short)TR::fromF(p[base + 3]) << 16);
+        unsigned w2 = (unsigned)(unsigned short)TR::fromF(p[base + 4])
+                      | ((unsigned)(unsigned short)TR::fromF(p[base + 5]) << 16);
+        unsigned w3 = (unsigned)(unsigned short)TR::fromF(p[base + 6])
+                      | ((unsigned)(unsigned short)TR::fromF(p[base + 7]) << 16);
+        auto s0 = __builtin_amdgcn_permlane32_swap((int)w0, (int)w2, false, false);
+        auto s1 = __builtin_amdgcn_permlane32_swap((int)w1, (int)w3, false, false);
+        intx4 pw;
+        pw[0] = s0[0];
+        pw[1] = s1[0];
+        pw[2] = s0[1];
+        pw[3] = s1[1];
+        pfrag[c] = *(frag*)&pw;
+      }
+
+      // ---- O^T += V^T P  (A = V^T from the swizzled image, B = P) ----
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          frag va = *(const frag*)vt_addr(Vb, mt * 32 + lo32,
+                                          (st * 32 + c * 16 + hi5 * 8) * 2, KS * 2);
+          oacc[mt] = MF::mma(va, pfrag[c], oacc[mt]);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+
+    if (kt + 1 < n_kt) write_tile((kt + 1) & 1);
+    __syncthreads();
+  }
+
+  // ---- epilogue: lane q = lo32 owns its whole row of O^T ----
+  if (qrow < S) {
+    const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+    const int64_t obase = (int64_t)b * o_sb + (int64_t)h * o_sh + (int64_t)qrow * o_sr;
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int d = mt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+        o[obase + d] = TR::fromF(oacc[mt][r] * inv_l);
+      }
+    if (hi5 == 0) lse[((int64_t)b * Hq + h) * S + qrow] = m_run + __logf(l_run);
   }
 }
 
@@ -686,15 +915,36 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 
 // ======================= C-ABI wrappers =======================
 
+static bool use_attn_v2() {
+  static int cached = -1;
+  if (cached < 0) {
+    const char* e = getenv("DK_ATTN_V2");
+    cached = (e && e[0] == '1') ? 1 : 0;
+  }
+  return cached == 1;
+}
+
 template <int DT, int D>
 static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
                            const void* v, int64_t B, int64_t Hq, int64_t Hkv,
                            int64_t S, float scale, int64_t o_sb, int64_t o_sh,
                            int64_t o_sr, dkStream stream) {
   using T = typename DTraits<DT>::T;
+  constexpr int KT = 64, KS = KT + 8, DS = D + 8;
+  if (!use_attn_v2()) {
+    // v3: swapped QK^T, in-register softmax, 128 q rows per workgroup
+    const int nQT = (int)((S + 127) / 128);
+    const int grid = (int)(B * Hq * nQT);
+    const size_t lds = sizeof(T) * (2 * KT * DS + 2 * D * KS);
+    hipLaunchKernelGGL((attn_fwd_v3_kernel<DT, D>), dim3(grid), dim3(256), lds,
+                       (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
+                       (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale,
+                       o_sb, o_sh, o_sr);
+    DK_CHECK_LAUNCH();
+    return 0;
+  }
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
-  constexpr int KT = 64, KS = KT + 8, DS = D + 8;
   const size_t lds = sizeof(T) * (2 * KT * DS + 2 * D * KS + 4 * 16 * KS);
   hipLaunchKernelGGL((attn_fwd_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
