@@ -410,22 +410,22 @@ __global__ __launch_bounds__(256) void attn_fwd_v3_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
 
-      // ---- in-lane online softmax for q row `qrow` ----
+      // ---- in-lane online softmax for q row `qrow` (p values live in the
+      // sc accumulator registers: no extra p[] array) ----
       const bool full_tile = (sbase + 32 <= q0) && (sbase + 32 <= S);
-      float p[16];
       float rm = NEG_BIG;
       if (full_tile) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          p[r] = sc[r] * scale;
-          rm = fmaxf(rm, p[r]);
+          sc[r] *= scale;
+          rm = fmaxf(rm, sc[r]);
         }
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int key = sbase + (r & 3) + 8 * (r >> 2) + 4 * hi5;
-          p[r] = (key > qrow || key >= S) ? NEG_BIG : sc[r] * scale;
-          rm = fmaxf(rm, p[r]);
+          sc[r] = (key > qrow || key >= S) ? NEG_BIG : sc[r] * scale;
+          rm = fmaxf(rm, sc[r]);
         }
       }
       rm = fmaxf(rm, __shfl_xor(rm, 32, DK_WAVE));
@@ -434,8 +434,8 @@ __global__ __launch_bounds__(256) void attn_fwd_v3_kernel(
       float psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        p[r] = p[r] <= NEG_BIG ? 0.f : __expf(p[r] - m_new);
-        psum += p[r];
+        sc[r] = sc[r] <= NEG_BIG ? 0.f : __expf(sc[r] - m_new);
+        psum += sc[r];
       }
       psum += __shfl_xor(psum, 32, DK_WAVE);
       l_run = l_run * alpha + psum;
@@ -450,18 +450,19 @@ __global__ __launch_bounds__(256) void attn_fwd_v3_kernel(
       // words are w0=(k0,k1) w1=(k2,k3) w2=(k8,k9) w3=(k10,k11) in the low
       // half (+4 in the high half); swap(w0,w2) -> frag dwords {d0, d2},
       // swap(w1,w3) -> {d1, d3}  (hardware-verified regrouping).
+      auto bits = [](T t) -> unsigned {
+        unsigned short u;
+        __builtin_memcpy(&u, &t, 2);
+        return (unsigned)u;
+      };
       frag pfrag[2];
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
         const int base = c * 8;
-        unsigned w0 = (unsigned)(unsigned short)TR::fromF(p[base + 0])
-                      | ((unsigned)(unsigned short)TR::fromF(p[base + 1]) << 16);
-        unsigned w1 = (unsigned)(unsigned short)TR::fromF(p[base + 2])
-                      | ((unsigned)(unsigned short)TR::fromF(p[base + 3]) << 16);
-        unsigned w2 = (unsigned)(unsigned short)TR::fromF(p[base + 4])
-                      | ((unsigned)(unsigned short)TR::fromF(p[base + 5]) << 16);
-        unsigned w3 = (unsigned)(unsigned short)TR::fromF(p[base + 6])
-                      | ((unsigned)(unsigned short)TR::fromF(p[base + 7]) << 16);
+        unsigned w0 = bits(TR::fromF(sc[base + 0])) | (bits(TR::fromF(sc[base + 1])) << 16);
+        unsigned w1 = bits(TR::fromF(sc[base + 2])) | (bits(TR::fromF(sc[base + 3])) << 16);
+        unsigned w2 = bits(TR::fromF(sc[base + 4])) | (bits(TR::fromF(sc[base + 5])) << 16);
+        unsigned w3 = bits(TR::fromF(sc[base + 6])) | (bits(TR::fromF(sc[base + 7])) << 16);
         auto s0 = __builtin_amdgcn_permlane32_swap((int)w0, (int)w2, false, false);
         auto s1 = __builtin_amdgcn_permlane32_swap((int)w1, (int)w3, false, false);
         intx4 pw;
