@@ -914,6 +914,392 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   }
 }
 
+// ======================= bwd v3: 32x32 MFMA tiles =======================
+// Same split (dK/dV over kv tiles, dQ over q tiles) and the same staged
+// images as v2, but on mfma_f32_32x32x16 tiles: a wave owns 32 keys (dkdv)
+// or 32 q rows (dq) and the workgroup covers 128, halving the MFMA / LDS
+// fragment-read instruction count per FLOP.  P^T / dS^T still re-shape
+// C->A through per-wave LDS tiles (a half-lane swap cannot transpose them).
+
+template <int DT, int D>
+__global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
+    typename DTraits<DT>::T* __restrict__ dk_out,
+    typename DTraits<DT>::T* __restrict__ dv_out,
+    const typename DTraits<DT>::T* __restrict__ do_,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA32<DT>;
+  using frag = typename MF::frag;
+  constexpr int QT = 32;
+  constexpr int QS = QT + 8;
+  constexpr int DS = D + 8;
+  constexpr int NKC = D / 16;
+  constexpr int NMT = D / 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* Q_lds = (T*)smem_raw;                   // [2][QT][DS]
+  T* QT_lds = Q_lds + 2 * QT * DS;           // [2][D][QS] swizzled (xmask 3)
+  T* dO_lds = QT_lds + 2 * D * QS;           // [2][QT][DS]
+  T* dOT_lds = dO_lds + 2 * QT * DS;         // [2][D][QS] swizzled
+  T* PT_lds = dOT_lds + 2 * D * QS;          // [4][32][QS]
+  T* DST_lds = PT_lds + 4 * 32 * QS;         // [4][32][QS]
+  float* lse_lds = (float*)(DST_lds + 4 * 32 * QS);  // [2][QT]
+  float* dl_lds = lse_lds + 2 * QT;                  // [2][QT]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo32 = lane & 31;
+  const int hi5 = lane >> 5;
+
+  const int nKT = (S + 127) / 128;
+  int bid = blockIdx.x;
+  const int kt = bid % nKT;
+  const int h = (bid / nKT) % Hq;
+  const int b = bid / (nKT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int k0 = kt * 128 + wave * 32;       // wave's first key
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+
+  // K,V A-fragments for this wave's keys (A row = key = lane&31)
+  frag k_frag[NKC], v_frag[NKC];
+  {
+    const int krow = k0 + lo32;
+    const int kr_c = krow < S ? krow : S - 1;
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      k_frag[kc] = *(const frag*)(k + kvoff + (int64_t)kr_c * D + kc * 16 + hi5 * 8);
+      v_frag[kc] = *(const frag*)(v + kvoff + (int64_t)kr_c * D + kc * 16 + hi5 * 8);
+    }
+  }
+
+  floatx16 dv_acc[NMT], dk_acc[NMT];
+#pragma unroll
+  for (int mt = 0; mt < NMT; ++mt) { dv_acc[mt] = (floatx16)(0.f); dk_acc[mt] = (floatx16)(0.f); }
+
+  const int qstart = (kt * 128) / QT;
+  const int nQT2 = (S + QT - 1) / QT;
+
+  const int st_t = (int)threadIdx.x;
+  const bool st_on = st_t < (QT * D) / 8;
+  const int st_row = st_t / (D / 8);
+  const int st_c8 = (st_t % (D / 8)) * 8;
+  shortx8 qreg, dreg;
+  float lse_reg = 0.f, dl_reg = 0.f;
+  const int64_t gbase = (int64_t)b * g_sb + (int64_t)h * g_sh;
+
+  auto load_qtile = [&](int qt2) {
+    const int qrow = qt2 * QT + st_row;
+    const int qr_c = qrow < S ? qrow : S - 1;
+    if (st_on) {
+      qreg = *(const shortx8*)(q + qoff + (int64_t)qr_c * D + st_c8);
+      dreg = *(const shortx8*)(do_ + gbase + (int64_t)qr_c * g_sr + st_c8);
+    }
+    if (st_t < QT) {
+      const int rr = qt2 * QT + st_t;
+      const int rr_c = rr < S ? rr : S - 1;
+      lse_reg = lse[lseoff + rr_c];
+      dl_reg = delta[lseoff + rr_c];
+    }
+  };
+  auto write_qtile = [&](int buf) {
+    if (st_on) {
+      *(shortx8*)(Q_lds + buf * QT * DS + st_row * DS + st_c8) = qreg;
+      *(shortx8*)(dO_lds + buf * QT * DS + st_row * DS + st_c8) = dreg;
+      T* QTb = QT_lds + buf * D * QS;
+      T* dOTb = dOT_lds + buf * D * QS;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        *vt_addr(QTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&qreg)[j];
+        *vt_addr(dOTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&dreg)[j];
+      }
+    }
+    if (st_t < QT) {
+      lse_lds[buf * QT + st_t] = lse_reg;
+      dl_lds[buf * QT + st_t] = dl_reg;
+    }
+  };
+
+  load_qtile(qstart);
+  write_qtile(0);
+  __syncthreads();
+
+  for (int qt2 = qstart; qt2 < nQT2; ++qt2) {
+    const int qbase = qt2 * QT;
+    const int cur = (qt2 - qstart) & 1;
+    T* Qb = Q_lds + cur * QT * DS;
+    T* dOb = dO_lds + cur * QT * DS;
+    T* QTb = QT_lds + cur * D * QS;
+    T* dOTb = dOT_lds + cur * D * QS;
+    const float* lse_b = lse_lds + cur * QT;
+    const float* dl_b = dl_lds + cur * QT;
+    if (qt2 + 1 < nQT2) load_qtile(qt2 + 1);
+
+    // ---- S^T = K Q^T and dP^T = V dO^T  (C: col = q = lane&31) ----
+    floatx16 st = (floatx16)(0.f), dpt = (floatx16)(0.f);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      frag bq = *(const frag*)(Qb + lo32 * DS + kc * 16 + hi5 * 8);
+      st = MF::mma(k_frag[kc], bq, st);
+      frag bd = *(const frag*)(dOb + lo32 * DS + kc * 16 + hi5 * 8);
+      dpt = MF::mma(v_frag[kc], bd, dpt);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    const int qcol = qbase + lo32;
+    const float lse_q = lse_b[lo32];
+    const float dl_q = dl_b[lo32];
+    const bool full = (qbase >= k0 + 32) && (qbase + QT <= S);
+    float pt[16], dst[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      float pv;
+      if (full) {
+        pv = __expf(st[r] * scale - lse_q);
+      } else {
+        const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+        pv = 0.f;
+        if (krow <= qcol && krow < S && qcol < S)
+          pv = __expf(st[r] * scale - lse_q);
+      }
+      pt[r] = pv;
+      dst[r] = pv * (dpt[r] - dl_q) * scale;
+    }
+
+    // ---- dV += P^T dO (A = P^T via LDS; B = dO_T) ----
+    T* Pw = PT_lds + wave * 32 * QS;
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      Pw[((r & 3) + 8 * (r >> 2) + 4 * hi5) * QS + lo32] = TR::fromF(pt[r]);
+    frag pa[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c)
+      pa[c] = *(const frag*)(Pw + lo32 * QS + c * 16 + hi5 * 8);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        frag bd = *(const frag*)vt_addr(dOTb, mt * 32 + lo32, (c * 16 + hi5 * 8) * 2, QS * 2, 3);
+        dv_acc[mt] = MF::mma(pa[c], bd, dv_acc[mt]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- dK += dS^T Q (A = dS^T via LDS; B = Q_T) ----
+    T* Dw = DST_lds + wave * 32 * QS;
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      Dw[((r & 3) + 8 * (r >> 2) + 4 * hi5) * QS + lo32] = TR::fromF(dst[r]);
+    frag da[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c)
+      da[c] = *(const frag*)(Dw + lo32 * QS + c * 16 + hi5 * 8);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        frag bq = *(const frag*)vt_addr(QTb, mt * 32 + lo32, (c * 16 + hi5 * 8) * 2, QS * 2, 3);
+        dk_acc[mt] = MF::mma(da[c], bq, dk_acc[mt]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (qt2 + 1 < nQT2) write_qtile(cur ^ 1);
+    __syncthreads();
+  }
+
+  // ---- write dK, dV (per q-head layout; caller sums GQA groups) ----
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+    if (krow >= S) continue;
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt) {
+      dk_out[qoff + (int64_t)krow * D + mt * 32 + lo32] = TR::fromF(dk_acc[mt][r]);
+      dv_out[qoff + (int64_t)krow * D + mt * 32 + lo32] = TR::fromF(dv_acc[mt][r]);
+    }
+  }
+}
+
+template <int DT, int D>
+__global__ __launch_bounds__(256) void attn_bwd_dq_v3_kernel(
+    typename DTraits<DT>::T* __restrict__ dq_out,
+    const typename DTraits<DT>::T* __restrict__ do_,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA32<DT>;
+  using frag = typename MF::frag;
+  constexpr int KT = 32;
+  constexpr int KS = KT + 8;
+  constexpr int DS = D + 8;
+  constexpr int NKC = D / 16;
+  constexpr int NMT = D / 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* K_lds = (T*)smem_raw;                    // [2][KT][DS]
+  T* V_lds = K_lds + 2 * KT * DS;             // [2][KT][DS]
+  T* KT_lds = V_lds + 2 * KT * DS;            // [2][D][KS] swizzled (xmask 3)
+  T* S_lds = KT_lds + 2 * D * KS;             // [4][32][KS]
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo32 = lane & 31;
+  const int hi5 = lane >> 5;
+
+  const int nQT = (S + 127) / 128;
+  int bid = blockIdx.x;
+  const int qt = bid % nQT;
+  const int h = (bid / nQT) % Hq;
+  const int b = bid / (nQT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int q0 = qt * 128 + wave * 32;
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+
+  // A-fragments (row = q = lane&31) + per-C-row lse/delta
+  frag q_frag[NKC], do_frag[NKC];
+  float lse_r[16], dl_r[16];
+  {
+    const int qrow = q0 + lo32;
+    const int qr_c = qrow < S ? qrow : S - 1;
+    const int64_t gb = (int64_t)b * g_sb + (int64_t)h * g_sh;
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      q_frag[kc] = *(const frag*)(q + qoff + (int64_t)qr_c * D + kc * 16 + hi5 * 8);
+      do_frag[kc] = *(const frag*)(do_ + gb + (int64_t)qr_c * g_sr + kc * 16 + hi5 * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int rr = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+      const int rr_c = rr < S ? rr : S - 1;
+      lse_r[r] = lse[lseoff + rr_c];
+      dl_r[r] = delta[lseoff + rr_c];
+    }
+  }
+
+  floatx16 dq_acc[NMT];
+#pragma unroll
+  for (int mt = 0; mt < NMT; ++mt) dq_acc[mt] = (floatx16)(0.f);
+
+  const int kv_end = min(S, qt * 128 + 128);
+  const int n_kt = (kv_end + KT - 1) / KT;
+
+  const int st_t = (int)threadIdx.x;
+  const bool st_on = st_t < (KT * D) / 8;
+  const int st_row = st_t / (D / 8);
+  const int st_c8 = (st_t % (D / 8)) * 8;
+  shortx8 kreg, vreg;
+
+  auto load_ktile = [&](int kt2) {
+    if (st_on) {
+      const int krow = kt2 * KT + st_row;
+      const int kr_c = krow < S ? krow : S - 1;
+      kreg = *(const shortx8*)(k + kvoff + (int64_t)kr_c * D + st_c8);
+      vreg = *(const shortx8*)(v + kvoff + (int64_t)kr_c * D + st_c8);
+    }
+  };
+  auto write_ktile = [&](int buf) {
+    if (st_on) {
+      *(shortx8*)(K_lds + buf * KT * DS + st_row * DS + st_c8) = kreg;
+      *(shortx8*)(V_lds + buf * KT * DS + st_row * DS + st_c8) = vreg;
+      T* KTb = KT_lds + buf * D * KS;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *vt_addr(KTb, st_c8 + j, st_row * 2, KS * 2, 3) = ((const T*)&kreg)[j];
+    }
+  };
+
+  load_ktile(0);
+  write_ktile(0);
+  __syncthreads();
+
+  for (int kt2 = 0; kt2 < n_kt; ++kt2) {
+    const int kbase = kt2 * KT;
+    const int cur = kt2 & 1;
+    T* Kb = K_lds + cur * KT * DS;
+    T* Vb = V_lds + cur * KT * DS;
+    T* KTb = KT_lds + cur * D * KS;
+    if (kt2 + 1 < n_kt) load_ktile(kt2 + 1);
+
+    // ---- S = Q K^T and dP = dO V^T  (C: col = key = lane&31) ----
+    floatx16 st = (floatx16)(0.f), dpt = (floatx16)(0.f);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      frag bk = *(const frag*)(Kb + lo32 * DS + kc * 16 + hi5 * 8);
+      st = MF::mma(q_frag[kc], bk, st);
+      frag bv = *(const frag*)(Vb + lo32 * DS + kc * 16 + hi5 * 8);
+      dpt = MF::mma(do_frag[kc], bv, dpt);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    const int kcol = kbase + lo32;
+    const bool full = (kbase + KT <= q0) && (kbase + KT <= S) && (q0 + 32 <= S);
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      float pv;
+      if (full) {
+        pv = __expf(st[r] * scale - lse_r[r]);
+      } else {
+        const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+        pv = 0.f;
+        if (kcol <= qrow && kcol < S && qrow < S)
+          pv = __expf(st[r] * scale - lse_r[r]);
+      }
+      ds[r] = pv * (dpt[r] - dl_r[r]) * scale;
+    }
+
+    // ---- dQ += dS K (A = dS via LDS; B = K_T) ----
+    T* Sw = S_lds + wave * 32 * KS;
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      Sw[((r & 3) + 8 * (r >> 2) + 4 * hi5) * KS + lo32] = TR::fromF(ds[r]);
+    frag da[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c)
+      da[c] = *(const frag*)(Sw + lo32 * KS + c * 16 + hi5 * 8);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        frag bk = *(const frag*)vt_addr(KTb, mt * 32 + lo32, (c * 16 + hi5 * 8) * 2, KS * 2, 3);
+        dq_acc[mt] = MF::mma(da[c], bk, dq_acc[mt]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (kt2 + 1 < n_kt) write_ktile(cur ^ 1);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+    if (qrow >= S) continue;
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+      dq_out[qoff + (int64_t)qrow * D + mt * 32 + lo32] = TR::fromF(dq_acc[mt][r]);
+  }
+}
+
 // ======================= C-ABI wrappers =======================
 
 static bool use_attn_v2() {
@@ -1001,9 +1387,21 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
                                 int64_t S, float scale, int64_t g_sb, int64_t g_sh,
                                 int64_t g_sr, dkStream stream) {
   using T = typename DTraits<DT>::T;
+  constexpr int QT = 32, QS = QT + 8, DS = D + 8;
+  if (!use_attn_v2()) {
+    const int nKT3 = (int)((S + 127) / 128);
+    const int grid3 = (int)(B * Hq * nKT3);
+    const size_t lds3 = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 4 * 32 * QS)
+                        + sizeof(float) * 4 * QT;
+    hipLaunchKernelGGL((attn_bwd_dkdv_v3_kernel<DT, D>), dim3(grid3), dim3(256), lds3,
+                       (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
+                       (const T*)q, (const T*)k, (const T*)v, lse, delta,
+                       (int)B, (int)Hq, (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
+    DK_CHECK_LAUNCH();
+    return 0;
+  }
   const int nKT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nKT);
-  constexpr int QT = 32, QS = QT + 8, DS = D + 8;
   const size_t lds = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 4 * 16 * QS) + sizeof(float) * 4 * QT;
   hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
@@ -1038,9 +1436,20 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
                               float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
                               dkStream stream) {
   using T = typename DTraits<DT>::T;
+  constexpr int KT = 32, KS = KT + 8, DS = D + 8;
+  if (!use_attn_v2()) {
+    const int nQT3 = (int)((S + 127) / 128);
+    const int grid3 = (int)(B * Hq * nQT3);
+    const size_t lds3 = sizeof(T) * (4 * KT * DS + 2 * D * KS + 4 * 32 * KS);
+    hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<DT, D>), dim3(grid3), dim3(256), lds3,
+                       (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
+                       (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
+                       (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
+    DK_CHECK_LAUNCH();
+    return 0;
+  }
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
-  constexpr int KT = 32, KS = KT + 8, DS = D + 8;
   const size_t lds = sizeof(T) * (4 * KT * DS + 2 * D * KS + 4 * 16 * KS);
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
