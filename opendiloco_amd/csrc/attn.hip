@@ -304,7 +304,7 @@ template <> struct MFMA32<1> {
 };
 
 template <int DT, int D>
-__global__ __launch_bounds__(256) void attn_fwd_v3_kernel(
+__global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
     typename DTraits<DT>::T* __restrict__ o, float* __restrict__ lse,
     const typename DTraits<DT>::T* __restrict__ q,
     const typename DTraits<DT>::T* __restrict__ k,
@@ -551,7 +551,7 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
 // LDS: Q[2][32][D+8] | Q_T[2][D][32+8] | dO[2][32][D+8] | dO_T[2][D][32+8] |
 //      lse[2][32] f32 | delta[2][32] f32 | P_T[4][16][32+8] | dS[4][16][32+8]
 template <int DT, int D>
-__global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
+__global__ __launch_bounds__(256, 3) void attn_bwd_dkdv_kernel(
     typename DTraits<DT>::T* __restrict__ dk_out,
     typename DTraits<DT>::T* __restrict__ dv_out,
     const typename DTraits<DT>::T* __restrict__ do_,
@@ -756,7 +756,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 // grid over (b, hq, q-tile of 64); wave owns 16 q rows.  Loop kv tiles of 32.
 // LDS: K[32][D+8] | V[32][D+8] | K_T[D][32+8] | dS[4][16][32+8]
 template <int DT, int D>
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     typename DTraits<DT>::T* __restrict__ dq_out,
     const typename DTraits<DT>::T* __restrict__ do_,
     const typename DTraits<DT>::T* __restrict__ q,
