@@ -969,17 +969,12 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
   const int64_t lseoff = ((int64_t)b * Hq + h) * S;
 
-  // K,V A-fragments for this wave's keys (A row = key = lane&31)
-  frag k_frag[NKC], v_frag[NKC];
-  {
-    const int krow = k0 + lo32;
-    const int kr_c = krow < S ? krow : S - 1;
-#pragma unroll
-    for (int kc = 0; kc < NKC; ++kc) {
-      k_frag[kc] = *(const frag*)(k + kvoff + (int64_t)kr_c * D + kc * 16 + hi5 * 8);
-      v_frag[kc] = *(const frag*)(v + kvoff + (int64_t)kr_c * D + kc * 16 + hi5 * 8);
-    }
-  }
+  // K,V A-fragment base (A row = key = lane&31); fragments are re-read per
+  // q-tile from L2 instead of held in registers — frees 32 VGPRs so the
+  // kernel fits 3 waves/SIMD without spills
+  const int kr_c0 = (k0 + lo32) < S ? (k0 + lo32) : S - 1;
+  const T* kbase_p = k + kvoff + (int64_t)kr_c0 * D + hi5 * 8;
+  const T* vbase_p = v + kvoff + (int64_t)kr_c0 * D + hi5 * 8;
 
   floatx16 dv_acc[NMT], dk_acc[NMT];
 #pragma unroll
@@ -1048,10 +1043,12 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
+      frag ka = *(const frag*)(kbase_p + kc * 16);
       frag bq = *(const frag*)(Qb + lo32 * DS + kc * 16 + hi5 * 8);
-      st = MF::mma(k_frag[kc], bq, st);
+      st = MF::mma(ka, bq, st);
+      frag va = *(const frag*)(vbase_p + kc * 16);
       frag bd = *(const frag*)(dOb + lo32 * DS + kc * 16 + hi5 * 8);
-      dpt = MF::mma(v_frag[kc], bd, dpt);
+      dpt = MF::mma(va, bd, dpt);
     }
     __builtin_amdgcn_s_setprio(0);
 
