@@ -37,6 +37,11 @@ CASES = {
                                           batch_size=16, per_device_train_batch_size=8,
                                           seq_length=1024, lr=1e-2, max_steps=10, seed=42,
                                           record_param_hash_every=5),
+    # longer horizon: 30 steps with 10 outer rounds (bf16-drift parity case)
+    "llama2m_w1_h3_long": OracleConfig(model_path=FIXTURE, n_workers=1, local_steps=3,
+                                       batch_size=16, per_device_train_batch_size=8,
+                                       seq_length=256, lr=1e-3, max_steps=30, seed=42,
+                                       record_param_hash_every=10),
 }
 
 
