@@ -551,7 +551,7 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
 // LDS: Q[2][32][D+8] | Q_T[2][D][32+8] | dO[2][32][D+8] | dO_T[2][D][32+8] |
 //      lse[2][32] f32 | delta[2][32] f32 | P_T[4][16][32+8] | dS[4][16][32+8]
 template <int DT, int D>
-__global__ __launch_bounds__(256, 3) void attn_bwd_dkdv_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     typename DTraits<DT>::T* __restrict__ dk_out,
     typename DTraits<DT>::T* __restrict__ dv_out,
     const typename DTraits<DT>::T* __restrict__ do_,
@@ -576,12 +576,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dkdv_kernel(
   T* QT_lds = Q_lds + 2 * QT * DS;           // [2][D][QS] swizzled
   T* dO_lds = QT_lds + 2 * D * QS;           // [2][QT][DS]
   T* dOT_lds = dO_lds + 2 * QT * DS;         // [2][D][QS] swizzled
-  T* PT_lds = dOT_lds + 2 * D * QS;          // [4][16][QS]  (P^T tiles)
-  T* DS_lds = PT_lds + 4 * 16 * QS;          // [4][16][QS]  (dS^T tiles; separate
+  T* PT_lds = dOT_lds + 2 * D * QS;          // [8][16][QS]  (P^T tiles)
+  T* DS_lds = PT_lds + 8 * 16 * QS;          // [8][16][QS]  (dS^T tiles; separate
                                              //  buffer: avoids an LDS WAR hazard
                                              //  between the P^T A-frag read and
                                              //  the dS^T writes in one iteration)
-  float* lse_lds = (float*)(DS_lds + 4 * 16 * QS);  // [2][QT]
+  float* lse_lds = (float*)(DS_lds + 8 * 16 * QS);  // [2][QT]
   float* dl_lds = lse_lds + 2 * QT;                 // [2][QT]
 
   const int lane = threadIdx.x & 63;
@@ -589,14 +589,14 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dkdv_kernel(
   const int lo = lane & 15;
   const int hi = lane >> 4;
 
-  const int nKT = (S + 63) / 64;
+  const int nKT = (S + 127) / 128;           // 8 waves x 16 keys per WG
   int bid = blockIdx.x;
   const int kt = bid % nKT;
   const int h = (bid / nKT) % Hq;
   const int b = bid / (nKT * Hq);
   const int hkv = h / (Hq / Hkv);
 
-  const int k0 = kt * 64 + wave * 16;        // this wave's first key
+  const int k0 = kt * 128 + wave * 16;       // this wave's first key
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
   const int64_t lseoff = ((int64_t)b * Hq + h) * S;
@@ -617,7 +617,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dkdv_kernel(
 #pragma unroll
   for (int dn = 0; dn < NDN; ++dn) { dv_acc[dn] = (floatx4)(0.f); dk_acc[dn] = (floatx4)(0.f); }
 
-  const int qstart = (kt * 64) / QT;         // first q tile that sees these keys
+  const int qstart = (kt * 128) / QT;        // first q tile that sees these keys
   const int nQT2 = (S + QT - 1) / QT;
 
   // async staging state (QT*D/8 <= 256 loads: one piece per thread)
@@ -756,7 +756,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dkdv_kernel(
 // grid over (b, hq, q-tile of 64); wave owns 16 q rows.  Loop kv tiles of 32.
 // LDS: K[32][D+8] | V[32][D+8] | K_T[D][32+8] | dS[4][16][32+8]
 template <int DT, int D>
-__global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     typename DTraits<DT>::T* __restrict__ dq_out,
     const typename DTraits<DT>::T* __restrict__ do_,
     const typename DTraits<DT>::T* __restrict__ q,
@@ -779,21 +779,21 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
   T* K_lds = (T*)smem_raw;                    // [2][KT][DS]
   T* V_lds = K_lds + 2 * KT * DS;             // [2][KT][DS]
   T* KT_lds = V_lds + 2 * KT * DS;            // [2][D][KS] swizzled (xmask 3)
-  T* S_lds = KT_lds + 2 * D * KS;             // [4][16][KS]
+  T* S_lds = KT_lds + 2 * D * KS;             // [8][16][KS]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int lo = lane & 15;
   const int hi = lane >> 4;
 
-  const int nQT = (S + 63) / 64;
+  const int nQT = (S + 127) / 128;            // 8 waves x 16 q rows per WG
   int bid = blockIdx.x;
   const int qt = bid % nQT;
   const int h = (bid / nQT) % Hq;
   const int b = bid / (nQT * Hq);
   const int hkv = h / (Hq / Hkv);
 
-  const int q0 = qt * 64 + wave * 16;
+  const int q0 = qt * 128 + wave * 16;
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
   const int64_t lseoff = ((int64_t)b * Hq + h) * S;
@@ -822,7 +822,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int dn = 0; dn < NDN; ++dn) dq_acc[dn] = (floatx4)(0.f);
 
-  const int kv_end = min(S, qt * 64 + 64);
+  const int kv_end = min(S, qt * 128 + 128);
   const int n_kt = (kv_end + KT - 1) / KT;
 
   // async double-buffered staging (one piece per thread: KT*D/8 <= 256)
@@ -1409,10 +1409,10 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
     DK_CHECK_LAUNCH();
     return 0;
   }
-  const int nKT = (int)((S + 63) / 64);
+  const int nKT = (int)((S + 127) / 128);    // 8-wave WG: 128 keys
   const int grid = (int)(B * Hq * nKT);
-  const size_t lds = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 4 * 16 * QS) + sizeof(float) * 4 * QT;
-  hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(256), lds,
+  const size_t lds = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 8 * 16 * QS) + sizeof(float) * 4 * QT;
+  hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
                      (const T*)q, (const T*)k, (const T*)v, lse, delta,
                      (int)B, (int)Hq, (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
@@ -1457,10 +1457,10 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
     DK_CHECK_LAUNCH();
     return 0;
   }
-  const int nQT = (int)((S + 63) / 64);
+  const int nQT = (int)((S + 127) / 128);    // 8-wave WG: 128 q rows
   const int grid = (int)(B * Hq * nQT);
-  const size_t lds = sizeof(T) * (4 * KT * DS + 2 * D * KS + 4 * 16 * KS);
-  hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(256), lds,
+  const size_t lds = sizeof(T) * (4 * KT * DS + 2 * D * KS + 8 * 16 * KS);
+  hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
                      (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
