@@ -400,6 +400,9 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
 #pragma unroll
     for (int st = 0; st < 2; ++st) {            // two 32-key subtiles
       const int sbase = kbase + st * 32;
+      // wave-uniform diagonal skip: keys all above this wave's q rows would
+      // be fully masked (rm = NEG_BIG -> alpha 1, psum 0: state unchanged).
+      if (sbase >= q0 + 32) continue;
       // ---- S^T = K Q^T ----
       floatx16 sc = (floatx16)(0.f);
       __builtin_amdgcn_s_setprio(1);
@@ -676,6 +679,10 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const float* dl_b = dl_lds + cur * QT;
     if (qt + 1 < nQT2) load_qtile(qt + 1);
 
+    // 8-wave WGs skew the diagonal: a wave whose 16 keys all sit above this
+    // q tile (k0 > every qcol) would compute an all-masked (zero) tile —
+    // skip the MFMAs (wave-uniform branch); staging + barrier still run.
+    if (qbase + QT > k0) {
     // ---- S^T = K Q^T (16 keys x 32 q), P^T = exp(scale*S^T - lse) ----
     float pt[2][4], dst[2][4];
     __builtin_amdgcn_s_setprio(1);
@@ -734,6 +741,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
       dk_acc[dn] = MF::mma(da, bq, dk_acc[dn]);
     }
     __builtin_amdgcn_s_setprio(0);
+    }  // end diagonal skip
 
     if (qt + 1 < nQT2) write_qtile(cur ^ 1);  // T14: write late
     __syncthreads();
@@ -863,6 +871,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     T* KTb = KT_lds + cur * D * KS;
     if (kt + 1 < n_kt) load_ktile(kt + 1);
 
+    // wave-uniform diagonal skip (8-wave skew): if every key in this kv
+    // tile exceeds this wave's last q row, the whole tile is masked to zero.
+    if (kbase < q0 + 16) {
     float ds[2][4];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -899,6 +910,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       dq_acc[dn] = MF::mma(da, bk, dq_acc[dn]);
     }
     __builtin_amdgcn_s_setprio(0);
+    }  // end diagonal skip
 
     if (kt + 1 < n_kt) write_ktile(cur ^ 1);  // T14: write late
     __syncthreads();
