@@ -107,10 +107,17 @@ int dk_cross_entropy_bwd(void* dlogits, const void* logits, const float* lse,
 /* o_sb/o_sh/o_sr (and g_* on the backward entries): stride triplet
  * (addr = b*sb + h*sh + s*sr) for the attention output / upstream gradient,
  * so o can be written directly in [B, S, Hq*D] and do read from it without
- * transpose copies; sb == 0 selects the contiguous [B,H,S,D] default. */
+ * transpose copies; sb == 0 selects the contiguous [B,H,S,D] default.
+ * v_sb/v_sh/v_sr (and dv_* on dk_attn_bwd_dkdv): the same triplet for V /
+ * its gradient, so V can be read straight out of the packed QKV projection
+ * [B, S, (Hq+2*Hkv)*D] and dV written back into the packed gradient without
+ * gather/scatter passes (V carries no RoPE rotation).  Rows must stay
+ * 16-byte aligned (strides in elements, last dim contiguous).  dv_* only
+ * makes sense when Hq == Hkv (no GQA group summation). */
 int dk_attn_fwd(void* o, float* lse, const void* q, const void* k, const void* v,
                 int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
                 float scale, int64_t o_sb, int64_t o_sh, int64_t o_sr,
+                int64_t v_sb, int64_t v_sh, int64_t v_sr,
                 int dtype, dkStream stream);
 /* delta[b,h,s] = rowsum(do * o), fp32 — preprocessing for bwd. */
 int dk_attn_bwd_preprocess(float* delta, const void* do_, const void* o,
@@ -122,11 +129,14 @@ int dk_attn_bwd_dkdv(void* dk_out, void* dv_out, const void* do_, const void* q,
                      const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
                      int64_t S, int64_t D, float scale,
                      int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                     int64_t v_sb, int64_t v_sh, int64_t v_sr,
+                     int64_t dv_sb, int64_t dv_sh, int64_t dv_sr,
                      int dtype, dkStream stream);
 int dk_attn_bwd_dq(void* dq_out, const void* do_, const void* q, const void* k,
                    const void* v, const float* lse, const float* delta,
                    int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
                    float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                   int64_t v_sb, int64_t v_sh, int64_t v_sr,
                    int dtype, dkStream stream);
 
 /* ---- Fused AdamW (inner optimizer) --------------------------------------

@@ -91,7 +91,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const typename DTraits<DT>::T* __restrict__ k,
     const typename DTraits<DT>::T* __restrict__ v,
     int B, int Hq, int Hkv, int S, float scale,
-    int64_t o_sb, int64_t o_sh, int64_t o_sr) {
+    int64_t o_sb, int64_t o_sh, int64_t o_sr,
+    int64_t v_sb, int64_t v_sh, int64_t v_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA16<DT>;
@@ -124,6 +125,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int q0 = qt * 64 + wave * 16;            // this wave's first q row
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t voff = v_sb ? ((int64_t)b * v_sb + (int64_t)hkv * v_sh) : kvoff;
+  const int64_t v_rs = v_sb ? v_sr : (int64_t)D;
 
   // Q A-fragments for this wave's 16 rows (clamped on the tail tile)
   frag q_frag[NKC];
@@ -156,7 +159,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       const int krow = kt * KT + st_row[i];
       const int kr_c = krow < S ? krow : S - 1;
       kreg[i] = *(const frag*)(k + kvoff + (int64_t)kr_c * D + st_c8[i]);
-      vreg[i] = *(const frag*)(v + kvoff + (int64_t)kr_c * D + st_c8[i]);
+      vreg[i] = *(const frag*)(v + voff + (int64_t)kr_c * v_rs + st_c8[i]);
     }
   };
   auto write_tile = [&](int buf) {
@@ -310,7 +313,8 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
     const typename DTraits<DT>::T* __restrict__ k,
     const typename DTraits<DT>::T* __restrict__ v,
     int B, int Hq, int Hkv, int S, float scale,
-    int64_t o_sb, int64_t o_sh, int64_t o_sr) {
+    int64_t o_sb, int64_t o_sh, int64_t o_sr,
+    int64_t v_sb, int64_t v_sh, int64_t v_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA32<DT>;
@@ -342,6 +346,8 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
   const int qrow = q0 + lo32;                    // this lane's q row
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t voff = v_sb ? ((int64_t)b * v_sb + (int64_t)hkv * v_sh) : kvoff;
+  const int64_t v_rs = v_sb ? v_sr : (int64_t)D;
 
   // persistent Q B-fragments: slot j of chunk kc = Q[qrow][kc*16 + hi5*8 + j]
   frag q_frag[NKC];
@@ -371,7 +377,7 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
       const int krow = kt * KT + st_row[i];
       const int kr_c = krow < S ? krow : S - 1;
       kreg[i] = *(const shortx8*)(k + kvoff + (int64_t)kr_c * D + st_c8[i]);
-      vreg[i] = *(const shortx8*)(v + kvoff + (int64_t)kr_c * D + st_c8[i]);
+      vreg[i] = *(const shortx8*)(v + voff + (int64_t)kr_c * v_rs + st_c8[i]);
     }
   };
   auto write_tile = [&](int buf) {
@@ -563,7 +569,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const typename DTraits<DT>::T* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
     int B, int Hq, int Hkv, int S, float scale,
-    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
+    int64_t g_sb, int64_t g_sh, int64_t g_sr,
+    int64_t v_sb, int64_t v_sh, int64_t v_sr,
+    int64_t dv_sb, int64_t dv_sh, int64_t dv_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA16<DT>;
@@ -603,6 +611,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
   const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+  const int64_t voff = v_sb ? ((int64_t)b * v_sb + (int64_t)hkv * v_sh) : kvoff;
+  const int64_t v_rs = v_sb ? v_sr : (int64_t)D;
 
   // K,V A-fragments for this wave's 16 keys
   frag k_frag[NKC], v_frag[NKC];
@@ -612,7 +622,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       k_frag[kc] = *(const frag*)(k + kvoff + (int64_t)kr_c * D + kc * 32 + hi * 8);
-      v_frag[kc] = *(const frag*)(v + kvoff + (int64_t)kr_c * D + kc * 32 + hi * 8);
+      v_frag[kc] = *(const frag*)(v + voff + (int64_t)kr_c * v_rs + kc * 32 + hi * 8);
     }
   }
 
@@ -748,6 +758,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   }
 
   // ---- write dK, dV (per q-head layout [B,Hq,S,D]; caller sums GQA groups) ----
+  const int64_t dvoff = dv_sb ? ((int64_t)b * dv_sb + (int64_t)h * dv_sh) : qoff;
+  const int64_t dv_rs = dv_sb ? dv_sr : (int64_t)D;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int krow = k0 + hi * 4 + r;
@@ -755,7 +767,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
       dk_out[qoff + (int64_t)krow * D + dn * 16 + lo] = TR::fromF(dk_acc[dn][r]);
-      dv_out[qoff + (int64_t)krow * D + dn * 16 + lo] = TR::fromF(dv_acc[dn][r]);
+      dv_out[dvoff + (int64_t)krow * dv_rs + dn * 16 + lo] = TR::fromF(dv_acc[dn][r]);
     }
   }
 }
@@ -772,7 +784,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const typename DTraits<DT>::T* __restrict__ v,
     const float* __restrict__ lse, const float* __restrict__ delta,
     int B, int Hq, int Hkv, int S, float scale,
-    int64_t g_sb, int64_t g_sh, int64_t g_sr) {
+    int64_t g_sb, int64_t g_sh, int64_t g_sr,
+    int64_t v_sb, int64_t v_sh, int64_t v_sr) {
   using TR = DTraits<DT>;
   using T = typename TR::T;
   using MF = MFMA16<DT>;
@@ -805,6 +818,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
   const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
   const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+  const int64_t voff = v_sb ? ((int64_t)b * v_sb + (int64_t)hkv * v_sh) : kvoff;
+  const int64_t v_rs = v_sb ? v_sr : (int64_t)D;
 
   frag q_frag[NKC], do_frag[NKC];
   float lse_r[4], dl_r[4];
@@ -845,7 +860,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       const int krow = kt * KT + st_row;
       const int kr_c = krow < S ? krow : S - 1;
       kreg = *(const frag*)(k + kvoff + (int64_t)kr_c * D + st_c8);
-      vreg = *(const frag*)(v + kvoff + (int64_t)kr_c * D + st_c8);
+      vreg = *(const frag*)(v + voff + (int64_t)kr_c * v_rs + st_c8);
     }
   };
   auto write_ktile = [&](int buf) {
@@ -1336,7 +1351,8 @@ template <int DT, int D>
 static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
                            const void* v, int64_t B, int64_t Hq, int64_t Hkv,
                            int64_t S, float scale, int64_t o_sb, int64_t o_sh,
-                           int64_t o_sr, dkStream stream) {
+                           int64_t o_sr, int64_t v_sb, int64_t v_sh, int64_t v_sr,
+                           dkStream stream) {
   using T = typename DTraits<DT>::T;
   constexpr int KT = 64, KS = KT + 8, DS = D + 8;
   if (!use_attn_v2()) {
@@ -1347,7 +1363,7 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
     hipLaunchKernelGGL((attn_fwd_v3_kernel<DT, D>), dim3(grid), dim3(256), lds,
                        (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
                        (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale,
-                       o_sb, o_sh, o_sr);
+                       o_sb, o_sh, o_sr, v_sb, v_sh, v_sr);
     DK_CHECK_LAUNCH();
     return 0;
   }
@@ -1357,7 +1373,7 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
   hipLaunchKernelGGL((attn_fwd_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
                      (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale,
-                     o_sb, o_sh, o_sr);
+                     o_sb, o_sh, o_sr, v_sb, v_sh, v_sr);
   DK_CHECK_LAUNCH();
   return 0;
 }
@@ -1365,16 +1381,17 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
 extern "C" int dk_attn_fwd(void* o, float* lse, const void* q, const void* k,
                            const void* v, int64_t B, int64_t Hq, int64_t Hkv,
                            int64_t S, int64_t D, float scale,
-                           int64_t o_sb, int64_t o_sh, int64_t o_sr, int dtype,
+                           int64_t o_sb, int64_t o_sh, int64_t o_sr,
+                           int64_t v_sb, int64_t v_sh, int64_t v_sr, int dtype,
                            dkStream stream) {
   if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
   if (o_sb == 0) { o_sb = Hq * S * D; o_sh = S * D; o_sr = D; }  // BHSD default
   if (D == 64) {
-    if (dtype == 2) return launch_attn_fwd<2, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
-    return launch_attn_fwd<1, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
+    if (dtype == 2) return launch_attn_fwd<2, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, v_sb, v_sh, v_sr, stream);
+    return launch_attn_fwd<1, 64>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, v_sb, v_sh, v_sr, stream);
   } else if (D == 32) {
-    if (dtype == 2) return launch_attn_fwd<2, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
-    return launch_attn_fwd<1, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, stream);
+    if (dtype == 2) return launch_attn_fwd<2, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, v_sb, v_sh, v_sr, stream);
+    return launch_attn_fwd<1, 32>(o, lse, q, k, v, B, Hq, Hkv, S, scale, o_sb, o_sh, o_sr, v_sb, v_sh, v_sr, stream);
   }
   return (int)hipErrorInvalidValue;
 }
@@ -1406,10 +1423,12 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
                                 const void* k, const void* v, const float* lse,
                                 const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
                                 int64_t S, float scale, int64_t g_sb, int64_t g_sh,
-                                int64_t g_sr, dkStream stream) {
+                                int64_t g_sr, int64_t v_sb, int64_t v_sh, int64_t v_sr,
+                                int64_t dv_sb, int64_t dv_sh, int64_t dv_sr,
+                                dkStream stream) {
   using T = typename DTraits<DT>::T;
   constexpr int QT = 32, QS = QT + 8, DS = D + 8;
-  if (use_bwd_v3()) {
+  if (use_bwd_v3() && v_sb == 0 && dv_sb == 0) {  // v3 port is contiguous-only
     const int nKT3 = (int)((S + 127) / 128);
     const int grid3 = (int)(B * Hq * nKT3);
     const size_t lds3 = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 4 * 32 * QS)
@@ -1427,7 +1446,8 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
   hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
                      (const T*)q, (const T*)k, (const T*)v, lse, delta,
-                     (int)B, (int)Hq, (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
+                     (int)B, (int)Hq, (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr,
+                     v_sb, v_sh, v_sr, dv_sb, dv_sh, dv_sr);
   DK_CHECK_LAUNCH();
   return 0;
 }
@@ -1436,16 +1456,18 @@ extern "C" int dk_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
                                 const void* k, const void* v, const float* lse,
                                 const float* delta, int64_t B, int64_t Hq, int64_t Hkv,
                                 int64_t S, int64_t D, float scale,
-                                int64_t g_sb, int64_t g_sh, int64_t g_sr, int dtype,
+                                int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                                int64_t v_sb, int64_t v_sh, int64_t v_sr,
+                                int64_t dv_sb, int64_t dv_sh, int64_t dv_sr, int dtype,
                                 dkStream stream) {
   if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
   if (g_sb == 0) { g_sb = Hq * S * D; g_sh = S * D; g_sr = D; }
   if (D == 64) {
-    if (dtype == 2) return launch_attn_bwd_dkdv<2, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
-    return launch_attn_bwd_dkdv<1, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    if (dtype == 2) return launch_attn_bwd_dkdv<2, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, dv_sb, dv_sh, dv_sr, stream);
+    return launch_attn_bwd_dkdv<1, 64>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, dv_sb, dv_sh, dv_sr, stream);
   } else if (D == 32) {
-    if (dtype == 2) return launch_attn_bwd_dkdv<2, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
-    return launch_attn_bwd_dkdv<1, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    if (dtype == 2) return launch_attn_bwd_dkdv<2, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, dv_sb, dv_sh, dv_sr, stream);
+    return launch_attn_bwd_dkdv<1, 32>(dk_o, dv_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, dv_sb, dv_sh, dv_sr, stream);
   }
   return (int)hipErrorInvalidValue;
 }
@@ -1455,10 +1477,11 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
                               const void* v, const float* lse, const float* delta,
                               int64_t B, int64_t Hq, int64_t Hkv, int64_t S,
                               float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                              int64_t v_sb, int64_t v_sh, int64_t v_sr,
                               dkStream stream) {
   using T = typename DTraits<DT>::T;
   constexpr int KT = 32, KS = KT + 8, DS = D + 8;
-  if (use_bwd_v3()) {
+  if (use_bwd_v3() && v_sb == 0) {  // v3 port is contiguous-only
     const int nQT3 = (int)((S + 127) / 128);
     const int grid3 = (int)(B * Hq * nQT3);
     const size_t lds3 = sizeof(T) * (4 * KT * DS + 2 * D * KS + 4 * 32 * KS);
@@ -1475,7 +1498,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
-                     (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr);
+                     (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr);
   DK_CHECK_LAUNCH();
   return 0;
 }
@@ -1484,15 +1507,16 @@ extern "C" int dk_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
                               const void* v, const float* lse, const float* delta,
                               int64_t B, int64_t Hq, int64_t Hkv, int64_t S, int64_t D,
                               float scale, int64_t g_sb, int64_t g_sh, int64_t g_sr,
+                              int64_t v_sb, int64_t v_sh, int64_t v_sr,
                               int dtype, dkStream stream) {
   if (dtype != 1 && dtype != 2) return (int)hipErrorInvalidValue;
   if (g_sb == 0) { g_sb = Hq * S * D; g_sh = S * D; g_sr = D; }
   if (D == 64) {
-    if (dtype == 2) return launch_attn_bwd_dq<2, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
-    return launch_attn_bwd_dq<1, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    if (dtype == 2) return launch_attn_bwd_dq<2, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, stream);
+    return launch_attn_bwd_dq<1, 64>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, stream);
   } else if (D == 32) {
-    if (dtype == 2) return launch_attn_bwd_dq<2, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
-    return launch_attn_bwd_dq<1, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, stream);
+    if (dtype == 2) return launch_attn_bwd_dq<2, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, stream);
+    return launch_attn_bwd_dq<1, 32>(dq_o, do_, q, k, v, lse, delta, B, Hq, Hkv, S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr, stream);
   }
   return (int)hipErrorInvalidValue;
 }

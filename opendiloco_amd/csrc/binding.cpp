@@ -169,7 +169,7 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
   auto o = at::empty_like(q);
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   DK_OK(dk_attn_fwd(o.data_ptr(), lse.data_ptr<float>(), q.data_ptr(), k.data_ptr(),
-                    v.data_ptr(), B, Hq, Hkv, S, D, (float)scale, 0, 0, 0,
+                    v.data_ptr(), B, Hq, Hkv, S, D, (float)scale, 0, 0, 0, 0, 0, 0,
                     dt_of(q), stream()));
   return {o, lse};
 }
@@ -190,10 +190,11 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& do_, const at::Tensor& q,
   DK_OK(dk_attn_bwd_dkdv(dk_full.data_ptr(), dv_full.data_ptr(), do_.data_ptr(),
                          q.data_ptr(), k.data_ptr(), v.data_ptr(), lse.data_ptr<float>(),
                          delta.data_ptr<float>(), B, Hq, Hkv, S, D, (float)scale,
-                         0, 0, 0, dt_of(q), stream()));
+                         0, 0, 0, 0, 0, 0, 0, 0, 0, dt_of(q), stream()));
   DK_OK(dk_attn_bwd_dq(dq.data_ptr(), do_.data_ptr(), q.data_ptr(), k.data_ptr(),
                        v.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       B, Hq, Hkv, S, D, (float)scale, 0, 0, 0, dt_of(q), stream()));
+                       B, Hq, Hkv, S, D, (float)scale, 0, 0, 0, 0, 0, 0,
+                       dt_of(q), stream()));
   return {dq, dk_full, dv_full};
 }
 
@@ -230,36 +231,64 @@ std::vector<at::Tensor> attn_fwd_bsd(const at::Tensor& q, const at::Tensor& k,
   CHECK_DEV_CONTIG(q);
   const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
   const int64_t Hkv = k.size(1);
+  // v may be a strided view into the packed QKV projection (last dim contiguous)
+  TORCH_CHECK(v.stride(3) == 1, "attn_fwd_bsd: v last dim must be contiguous");
+  const bool vc = v.is_contiguous();
   auto o = at::empty({B, S, Hq * D}, q.options());
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   DK_OK(dk_attn_fwd(o.data_ptr(), lse.data_ptr<float>(), q.data_ptr(), k.data_ptr(),
                     v.data_ptr(), B, Hq, Hkv, S, D, (float)scale,
-                    /*o strides (b,h,s)*/ S * Hq * D, D, Hq * D, dt_of(q), stream()));
+                    /*o strides (b,h,s)*/ S * Hq * D, D, Hq * D,
+                    vc ? 0 : v.stride(0), vc ? 0 : v.stride(1), vc ? 0 : v.stride(2),
+                    dt_of(q), stream()));
   return {o, lse};
 }
 
 std::vector<at::Tensor> attn_bwd_bsd(const at::Tensor& do_bsd, const at::Tensor& q,
                                      const at::Tensor& k, const at::Tensor& v,
                                      const at::Tensor& o_bsd, const at::Tensor& lse,
-                                     double scale) {
+                                     double scale,
+                                     c10::optional<at::Tensor> dv_out = c10::nullopt) {
   CHECK_DEV_CONTIG(do_bsd);
   const int64_t B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
   const int64_t Hkv = k.size(1);
   const int64_t sb = S * Hq * D, sh = D, sr = Hq * D;
+  TORCH_CHECK(v.stride(3) == 1, "attn_bwd_bsd: v last dim must be contiguous");
+  const bool vc = v.is_contiguous();
+  const int64_t v_sb = vc ? 0 : v.stride(0), v_sh = vc ? 0 : v.stride(1),
+                v_sr = vc ? 0 : v.stride(2);
   auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   DK_OK(dk_attn_bwd_preprocess(delta.data_ptr<float>(), do_bsd.data_ptr(),
                                o_bsd.data_ptr(), B, Hq, S, D, sb, sh, sr,
                                dt_of(q), stream()));
   auto dq = at::empty_like(q);
   auto dk_full = at::empty({B, Hq, S, D}, q.options());
-  auto dv_full = at::empty({B, Hq, S, D}, q.options());
-  DK_OK(dk_attn_bwd_dkdv(dk_full.data_ptr(), dv_full.data_ptr(), do_bsd.data_ptr(),
+  at::Tensor dv_full;
+  int64_t dv_sb = 0, dv_sh = 0, dv_sr = 0;
+  void* dv_ptr;
+  if (dv_out.has_value()) {
+    // write dV straight into a strided view (e.g. the packed dQKV buffer);
+    // only valid without GQA group summation
+    TORCH_CHECK(Hq == Hkv, "attn_bwd_bsd: direct dv_out requires Hq == Hkv");
+    TORCH_CHECK(dv_out->stride(3) == 1, "attn_bwd_bsd: dv_out last dim must be contiguous");
+    dv_full = *dv_out;
+    dv_sb = dv_full.stride(0);
+    dv_sh = dv_full.stride(1);
+    dv_sr = dv_full.stride(2);
+    dv_ptr = dv_full.data_ptr();
+  } else {
+    dv_full = at::empty({B, Hq, S, D}, q.options());
+    dv_ptr = dv_full.data_ptr();
+  }
+  DK_OK(dk_attn_bwd_dkdv(dk_full.data_ptr(), dv_ptr, do_bsd.data_ptr(),
                          q.data_ptr(), k.data_ptr(), v.data_ptr(), lse.data_ptr<float>(),
                          delta.data_ptr<float>(), B, Hq, Hkv, S, D, (float)scale,
-                         sb, sh, sr, dt_of(q), stream()));
+                         sb, sh, sr, v_sb, v_sh, v_sr, dv_sb, dv_sh, dv_sr,
+                         dt_of(q), stream()));
   DK_OK(dk_attn_bwd_dq(dq.data_ptr(), do_bsd.data_ptr(), q.data_ptr(), k.data_ptr(),
                        v.data_ptr(), lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       B, Hq, Hkv, S, D, (float)scale, sb, sh, sr, dt_of(q), stream()));
+                       B, Hq, Hkv, S, D, (float)scale, sb, sh, sr, v_sb, v_sh, v_sr,
+                       dt_of(q), stream()));
   return {dq, dk_full, dv_full};
 }
 
@@ -364,7 +393,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_gather", &qkv_rope_gather);
   m.def("rope_scatter_", &rope_scatter_);
   m.def("attn_fwd_bsd", &attn_fwd_bsd);
-  m.def("attn_bwd_bsd", &attn_bwd_bsd);
+  m.def("attn_bwd_bsd", &attn_bwd_bsd, py::arg("do_bsd"), py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("o_bsd"), py::arg("lse"), py::arg("scale"),
+        py::arg("dv_out") = py::none());
   m.def("fused_adamw", &fused_adamw);
   m.def("clip_grad_", &clip_grad_);
   m.def("grad_norm", &grad_norm);

@@ -355,44 +355,58 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float) -
     return _AttentionFn.apply(q, k, v, scale)
 
 
+def _v_view(qkv, Hq, Hkv, D):
+    """[B,Hkv,S,D] strided view of the V columns of the packed [B,S,(Hq+2Hkv)D]
+    buffer (no copy; V carries no RoPE rotation so the attention kernels read
+    it in place via the v-stride triplet of the C ABI)."""
+    B, S, _ = qkv.shape
+    return qkv.view(B, S, Hq + 2 * Hkv, D).narrow(2, Hq + Hkv, Hkv).permute(0, 2, 1, 3)
+
+
 class _QKVRopeAttentionFn(torch.autograd.Function):
     """GPU-only fused path over the packed QKV projection output
-    [B, S, (Hq+2*Hkv)*D]: RoPE-gather q/k (+ plain gather v) straight out of
-    the packed buffer into [B,H,S,D], flash attention writing o directly as
-    [B, S, Hq*D], and a backward that RoPE-scatters dq/dk/dv back into ONE
-    dqkv buffer — no transpose+contiguous copies, no split/cat."""
+    [B, S, (Hq+2*Hkv)*D]: RoPE-gather q/k straight out of the packed buffer
+    into [B,H,S,D] while V is read IN PLACE through a strided view, flash
+    attention writing o directly as [B, S, Hq*D], and a backward that
+    RoPE-scatters dq/dk into ONE dqkv buffer with dV written directly into
+    its V columns (no GQA) — no transpose+contiguous copies, no split/cat,
+    no V gather/scatter passes."""
 
     @staticmethod
     def forward(ctx, qkv, cos, sin, Hq, Hkv, D, scale):
         ext = _ext()
-        nq, nkv = Hq * D, Hkv * D
+        nq = Hq * D
         q = ext.qkv_rope_gather(qkv, cos, sin, Hq, D, 0, True)
         k = ext.qkv_rope_gather(qkv, cos, sin, Hkv, D, nq, True)
-        v = ext.qkv_rope_gather(qkv, cos, sin, Hkv, D, nq + nkv, False)
+        v = _v_view(qkv, Hq, Hkv, D)
         o_bsd, lse = ext.attn_fwd_bsd(q, k, v, scale)
-        ctx.save_for_backward(q, k, v, o_bsd, lse, cos, sin)
+        ctx.save_for_backward(q, k, qkv, o_bsd, lse, cos, sin)
         ctx.dims = (Hq, Hkv, D, scale)
         return o_bsd
 
     @staticmethod
     def backward(ctx, do_bsd):
-        q, k, v, o_bsd, lse, cos, sin = ctx.saved_tensors
+        q, k, qkv, o_bsd, lse, cos, sin = ctx.saved_tensors
         Hq, Hkv, D, scale = ctx.dims
         ext = _ext()
-        dq, dkf, dvf = ext.attn_bwd_bsd(do_bsd.contiguous(), q, k, v, o_bsd, lse, scale)
-        if Hq != Hkv:
-            g = Hq // Hkv
-            B, _, S, _ = q.shape
-            dk = dkf.view(B, Hkv, g, S, D).sum(2).contiguous()
-            dv = dvf.view(B, Hkv, g, S, D).sum(2).contiguous()
-        else:
-            dk, dv = dkf, dvf
+        v = _v_view(qkv, Hq, Hkv, D)
         B, _, S, _ = q.shape
         nq, nkv = Hq * D, Hkv * D
         dqkv = torch.empty(B, S, nq + 2 * nkv, dtype=q.dtype, device=q.device)
+        if Hq == Hkv:
+            dv_out = _v_view(dqkv, Hq, Hkv, D)
+            dq, dkf, _ = ext.attn_bwd_bsd(do_bsd.contiguous(), q, k, v, o_bsd, lse,
+                                          scale, dv_out)
+            dk = dkf
+        else:
+            dq, dkf, dvf = ext.attn_bwd_bsd(do_bsd.contiguous(), q, k, v, o_bsd, lse,
+                                            scale)
+            g = Hq // Hkv
+            dk = dkf.view(B, Hkv, g, S, D).sum(2).contiguous()
+            dv = dvf.view(B, Hkv, g, S, D).sum(2).contiguous()
+            ext.rope_scatter_(dqkv, dv, cos, sin, nq + nkv, False)
         ext.rope_scatter_(dqkv, dq, cos, sin, 0, True)
         ext.rope_scatter_(dqkv, dk, cos, sin, nq, True)
-        ext.rope_scatter_(dqkv, dv, cos, sin, nq + nkv, False)
         return dqkv, None, None, None, None, None, None
 
 
