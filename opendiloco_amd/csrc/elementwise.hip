@@ -100,6 +100,66 @@ __global__ void rmsnorm_fwd_kernel(typename DTraits<DT>::T* __restrict__ y,
   }
 }
 
+// vectorized forward for the hot hidden sizes (cols == NT*VEC): thread t
+// owns columns [t*VEC, t*VEC+VEC) so the whole row lives in registers
+// between the sum-of-squares pass and the normalize pass — x/res are read
+// once, h/y written once, no strided second read (fwd 4.2 -> ~6 TB/s).
+template <int DT, int NT, int VEC, bool RES>
+__global__ void rmsnorm_fwd_vec_kernel(typename DTraits<DT>::T* __restrict__ y,
+                                       typename DTraits<DT>::T* __restrict__ h_out,
+                                       float* __restrict__ invrms,
+                                       const typename DTraits<DT>::T* __restrict__ x,
+                                       const typename DTraits<DT>::T* __restrict__ res,
+                                       const typename DTraits<DT>::T* __restrict__ w,
+                                       int64_t rows, int cols, float eps) {
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  typedef __attribute__((ext_vector_type(VEC))) short vec_t;
+  __shared__ float sred[NT / DK_WAVE];
+  const int c0 = threadIdx.x * VEC;
+  float wf[VEC];
+  {
+    vec_t wvv = *(const vec_t*)(w + c0);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) wf[j] = TR::toF(((const T*)&wvv)[j]);
+  }
+  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
+    float hf[VEC], ss = 0.f;
+    {
+      vec_t xv = *(const vec_t*)(x + r * cols + c0);
+      if (RES) {
+        vec_t rv = *(const vec_t*)(res + r * cols + c0);
+        vec_t hv;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          T h = TR::fromF(TR::toF(((const T*)&xv)[j]) + TR::toF(((const T*)&rv)[j]));
+          ((T*)&hv)[j] = h;
+          hf[j] = TR::toF(h);
+          ss += hf[j] * hf[j];
+        }
+        *(vec_t*)(h_out + r * cols + c0) = hv;
+      } else {
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          hf[j] = TR::toF(((const T*)&xv)[j]);
+          ss += hf[j] * hf[j];
+        }
+      }
+    }
+    ss = block_reduce_sum<NT>(ss, sred);
+    const float ir = rsqrtf(ss / (float)cols + eps);
+    if (threadIdx.x == 0 && invrms) invrms[r] = ir;
+    vec_t yv;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      // mirror HF order: xhat rounded to T first, then multiplied by w
+      float xh = TR::toF(TR::fromF(hf[j] * ir));
+      ((T*)&yv)[j] = TR::fromF(wf[j] * xh);
+    }
+    *(vec_t*)(y + r * cols + c0) = yv;
+  }
+}
+
 // backward: let xhat = x*invrms, g = dy*w.
 //   dx = (g - xhat * mean(g*xhat)) * invrms
 //   dw partials accumulate in REGISTERS per block (thread t owns columns
@@ -249,6 +309,31 @@ extern "C" int dk_rmsnorm_fwd(void* y, void* h_out, float* invrms, const void* x
   int grid = (int)(rows < 4096 ? rows : 4096);
   DK_DISPATCH_DT(dtype, {
     using T = typename DTraits<kDT>::T;
+    if constexpr (kDT != 0) {
+      if (cols == NT * 4 || cols == NT * 8) {  // vectorized: row in registers
+        if (cols == NT * 4) {
+          if (res != nullptr)
+            hipLaunchKernelGGL((rmsnorm_fwd_vec_kernel<kDT, NT, 4, true>), dim3(grid), dim3(NT), 0,
+                               (hipStream_t)stream, (T*)y, (T*)h_out, invrms, (const T*)x,
+                               (const T*)res, (const T*)w, rows, (int)cols, eps);
+          else
+            hipLaunchKernelGGL((rmsnorm_fwd_vec_kernel<kDT, NT, 4, false>), dim3(grid), dim3(NT), 0,
+                               (hipStream_t)stream, (T*)y, nullptr, invrms, (const T*)x,
+                               nullptr, (const T*)w, rows, (int)cols, eps);
+        } else {
+          if (res != nullptr)
+            hipLaunchKernelGGL((rmsnorm_fwd_vec_kernel<kDT, NT, 8, true>), dim3(grid), dim3(NT), 0,
+                               (hipStream_t)stream, (T*)y, (T*)h_out, invrms, (const T*)x,
+                               (const T*)res, (const T*)w, rows, (int)cols, eps);
+          else
+            hipLaunchKernelGGL((rmsnorm_fwd_vec_kernel<kDT, NT, 8, false>), dim3(grid), dim3(NT), 0,
+                               (hipStream_t)stream, (T*)y, nullptr, invrms, (const T*)x,
+                               nullptr, (const T*)w, rows, (int)cols, eps);
+        }
+        hipError_t e = hipGetLastError();
+        return (int)e;
+      }
+    }
     if (res != nullptr)
       hipLaunchKernelGGL((rmsnorm_fwd_kernel<kDT, NT, true>), dim3(grid), dim3(NT), 0,
                          (hipStream_t)stream, (T*)y, (T*)h_out, invrms, (const T*)x,
