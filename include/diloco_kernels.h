@@ -176,6 +176,11 @@ int dk_outer_nesterov(float* theta_outer, float* theta_local, float* momentum_bu
 /* ---- misc ---------------------------------------------------------------
  * Elementwise scaled cast between f32 and bf16/f16 flat buffers (used for
  * fp16/bf16 all-reduce payloads when compression is requested). */
+/* fp32 dst[i] += toF(src[i]) — fused low-precision dW accumulation into the
+ * fp32 master gradient (replaces the reference's autocast cast+add pair,
+ * train_diloco_torch.py:305-310 grad flow). */
+int dk_accum(float* dst, const void* src, int64_t n, int src_dtype,
+             dkStream stream);
 int dk_cast(void* dst, const void* src, int64_t n, int dst_dtype, int src_dtype,
             dkStream stream);
 

@@ -292,6 +292,16 @@ std::vector<at::Tensor> attn_bwd_bsd(const at::Tensor& do_bsd, const at::Tensor&
   return {dq, dk_full, dv_full};
 }
 
+// fp32 master-grad += low-precision dW (fused cast+accumulate)
+void accum_(at::Tensor& dst, const at::Tensor& src) {
+  CHECK_DEV_CONTIG(dst);
+  CHECK_DEV_CONTIG(src);
+  TORCH_CHECK(dst.scalar_type() == at::kFloat, "accum_: dst must be fp32");
+  TORCH_CHECK(dst.numel() == src.numel(), "accum_: numel mismatch");
+  DK_OK(dk_accum(dst.data_ptr<float>(), src.data_ptr(), dst.numel(),
+                 dt_of(src), stream()));
+}
+
 // ---- optimizer / outer step ----
 void fused_adamw(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
                  double lr, double beta1, double beta2, double eps, double weight_decay,
@@ -393,6 +403,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_gather", &qkv_rope_gather);
   m.def("rope_scatter_", &rope_scatter_);
   m.def("attn_fwd_bsd", &attn_fwd_bsd);
+  m.def("accum_", &accum_);
   m.def("attn_bwd_bsd", &attn_bwd_bsd, py::arg("do_bsd"), py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("o_bsd"), py::arg("lse"), py::arg("scale"),
         py::arg("dv_out") = py::none());

@@ -58,14 +58,20 @@ class _CastLinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w_master, w16):
-        ctx.save_for_backward(x, w16)
+        ctx.save_for_backward(x, w16, w_master)
         return F.linear(x, w16)
 
     @staticmethod
     def backward(ctx, dy):
-        x, w16 = ctx.saved_tensors
+        x, w16, w_master = ctx.saved_tensors
         dx = dy @ w16
         dw = (dy.reshape(-1, dy.shape[-1]).t() @ x.reshape(-1, x.shape[-1]))
+        wg = w_master.grad
+        if wg is not None and wg.is_cuda and dw.is_cuda:
+            # fused fp32 += bf16 into the flat master grad: one kernel per
+            # weight instead of the cast + autograd-add pair
+            ops._ext().accum_(wg.reshape(-1), dw.reshape(-1))
+            return dx, None, None
         return dx, dw.to(torch.float32), None
 
 
@@ -76,15 +82,19 @@ class _CastEmbeddingFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, ids, w_master, w16):
-        ctx.save_for_backward(ids)
+        ctx.save_for_backward(ids, w_master)
         ctx.num_weights = w_master.shape[0]
         return F.embedding(ids, w16)
 
     @staticmethod
     def backward(ctx, dy):
-        (ids,) = ctx.saved_tensors
+        ids, w_master = ctx.saved_tensors
         dw = torch.ops.aten.embedding_dense_backward(
             dy, ids, ctx.num_weights, -1, False)
+        wg = w_master.grad
+        if wg is not None and wg.is_cuda and dw.is_cuda:
+            ops._ext().accum_(wg.reshape(-1), dw.reshape(-1))
+            return None, None, None
         return None, dw.to(torch.float32), None
 
 
@@ -122,15 +132,22 @@ class _MultiCastLinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w16cat, sizes, *masters):
-        ctx.save_for_backward(x, w16cat)
+        ctx.save_for_backward(x, w16cat, *masters)
         ctx.sizes = sizes
         return F.linear(x, w16cat)
 
     @staticmethod
     def backward(ctx, dy):
-        x, w16cat = ctx.saved_tensors
+        x, w16cat, *masters = ctx.saved_tensors
         dx = dy @ w16cat
         dwcat = dy.reshape(-1, dy.shape[-1]).t() @ x.reshape(-1, x.shape[-1])
+        if dwcat.is_cuda and all(m.grad is not None and m.grad.is_cuda for m in masters):
+            ext = ops._ext()
+            off = 0
+            for n, m in zip(ctx.sizes, masters):
+                ext.accum_(m.grad.reshape(-1), dwcat[off:off + n].reshape(-1))
+                off += n
+            return (dx, None, None, *([None] * len(masters)))
         dws = []
         off = 0
         for n in ctx.sizes:

@@ -302,6 +302,19 @@ def test_pseudo_grad_and_nesterov(ext):
 
 
 @requires_gpu
+def test_accum(ext):
+    """accum_ must equal the cast + add pair it replaces, bit-exactly."""
+    torch.manual_seed(7)
+    for n in (64, 2688 * 1024, 2688 * 1024 + 5):
+        dst = torch.randn(n, device="cuda")
+        ref = dst.clone()
+        src = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+        ext.accum_(dst, src)
+        ref += src.to(torch.float32)
+        assert torch.equal(dst, ref), f"n={n}"
+
+
+@requires_gpu
 def test_cast(ext):
     src = torch.randn(12345, device="cuda")
     dst = torch.empty(12345, device="cuda", dtype=torch.bfloat16)
