@@ -43,12 +43,19 @@ __global__ void ce_fwd_kernel(float* __restrict__ loss_rows, float* __restrict__
     float m = -INFINITY, s = 0.f;
     for (int64_t i = threadIdx.x; i < nvec; i += NT) {
       V8 xv = *(const V8*)(row + i * W);
+      // per-vector max first, then one rescale + W independent exps: breaks
+      // the per-element serial rescale chain (latency-bound) into ILP
+      float xf[W], vm = -INFINITY;
 #pragma unroll
       for (int j = 0; j < W; ++j) {
-        float x = TR::toF(((const TT*)&xv)[j]);
-        if (x > m) { s *= __expf(m - x); m = x; }
-        s += __expf(x - m);
+        xf[j] = TR::toF(((const TT*)&xv)[j]);
+        vm = fmaxf(vm, xf[j]);
       }
+      if (vm > m) { s *= __expf(m - vm); m = vm; }
+      float ps = 0.f;
+#pragma unroll
+      for (int j = 0; j < W; ++j) ps += __expf(xf[j] - m);
+      s += ps;
     }
     for (int64_t c = nvec * W + threadIdx.x; c < V; c += NT) {
       float x = TR::toF(row[c]);
