@@ -123,28 +123,35 @@ __global__ void rmsnorm_fwd_vec_kernel(typename DTraits<DT>::T* __restrict__ y,
 #pragma unroll
     for (int j = 0; j < VEC; ++j) wf[j] = TR::toF(((const T*)&wvv)[j]);
   }
+  // software pipeline: next row's loads are issued before this row's
+  // block-reduce barrier so HBM latency overlaps the sync
+  vec_t xv, rv;
+  if (blockIdx.x < rows) {
+    xv = *(const vec_t*)(x + blockIdx.x * cols + c0);
+    if (RES) rv = *(const vec_t*)(res + blockIdx.x * cols + c0);
+  }
   for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
     float hf[VEC], ss = 0.f;
-    {
-      vec_t xv = *(const vec_t*)(x + r * cols + c0);
-      if (RES) {
-        vec_t rv = *(const vec_t*)(res + r * cols + c0);
-        vec_t hv;
+    if (RES) {
+      vec_t hv;
 #pragma unroll
-        for (int j = 0; j < VEC; ++j) {
-          T h = TR::fromF(TR::toF(((const T*)&xv)[j]) + TR::toF(((const T*)&rv)[j]));
-          ((T*)&hv)[j] = h;
-          hf[j] = TR::toF(h);
-          ss += hf[j] * hf[j];
-        }
-        *(vec_t*)(h_out + r * cols + c0) = hv;
-      } else {
-#pragma unroll
-        for (int j = 0; j < VEC; ++j) {
-          hf[j] = TR::toF(((const T*)&xv)[j]);
-          ss += hf[j] * hf[j];
-        }
+      for (int j = 0; j < VEC; ++j) {
+        T h = TR::fromF(TR::toF(((const T*)&xv)[j]) + TR::toF(((const T*)&rv)[j]));
+        ((T*)&hv)[j] = h;
+        hf[j] = TR::toF(h);
+        ss += hf[j] * hf[j];
       }
+      *(vec_t*)(h_out + r * cols + c0) = hv;
+    } else {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        hf[j] = TR::toF(((const T*)&xv)[j]);
+        ss += hf[j] * hf[j];
+      }
+    }
+    if (r + gridDim.x < rows) {
+      xv = *(const vec_t*)(x + (r + gridDim.x) * cols + c0);
+      if (RES) rv = *(const vec_t*)(res + (r + gridDim.x) * cols + c0);
     }
     ss = block_reduce_sum<NT>(ss, sred);
     const float ir = rsqrtf(ss / (float)cols + eps);
@@ -265,17 +272,25 @@ __global__ void rmsnorm_bwd_vec_kernel(typename DTraits<DT>::T* __restrict__ dx,
 #pragma unroll
     for (int j = 0; j < VEC; ++j) { wv[j] = TR::toF(((const T*)&wvv)[j]); dwacc[j] = 0.f; }
   }
+  vec_t xv, dv, rv;
+  if (blockIdx.x < rows) {
+    xv = *(const vec_t*)(x + blockIdx.x * cols + c0);
+    dv = *(const vec_t*)(dy + blockIdx.x * cols + c0);
+    if (DRES) rv = *(const vec_t*)(dres + blockIdx.x * cols + c0);
+  }
   for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
     const float ir = invrms[r];
-    float xf[VEC], dyf[VEC];
-    {
-      vec_t xv = *(const vec_t*)(x + r * cols + c0);
-      vec_t dv = *(const vec_t*)(dy + r * cols + c0);
+    float xf[VEC], dyf[VEC], drf[VEC];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        xf[j] = TR::toF(((const T*)&xv)[j]);
-        dyf[j] = TR::toF(((const T*)&dv)[j]);
-      }
+    for (int j = 0; j < VEC; ++j) {
+      xf[j] = TR::toF(((const T*)&xv)[j]);
+      dyf[j] = TR::toF(((const T*)&dv)[j]);
+      if (DRES) drf[j] = TR::toF(((const T*)&rv)[j]);
+    }
+    if (r + gridDim.x < rows) {
+      xv = *(const vec_t*)(x + (r + gridDim.x) * cols + c0);
+      dv = *(const vec_t*)(dy + (r + gridDim.x) * cols + c0);
+      if (DRES) rv = *(const vec_t*)(dres + (r + gridDim.x) * cols + c0);
     }
     float s1 = 0.f;
 #pragma unroll
@@ -287,7 +302,7 @@ __global__ void rmsnorm_bwd_vec_kernel(typename DTraits<DT>::T* __restrict__ dx,
       float xh = xf[j] * ir;
       float g = dyf[j] * wv[j];
       float dv = (g - xh * dot) * ir;
-      if (DRES) dv += TR::toF(dres[r * cols + c0 + j]);
+      if (DRES) dv += drf[j];
       ((T*)&dxv)[j] = TR::fromF(dv);
       dwacc[j] += dyf[j] * xh;
     }
