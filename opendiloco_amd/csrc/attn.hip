@@ -48,6 +48,44 @@ template <> struct MFMA16<1> {
   }
 };
 
+// gfx950 hardware transpose-read: within each 16-lane group, the lanes
+// supply the 8-byte chunks of a [4 rows][16 cols] bf16/f16 tile and lane L
+// receives column L&15: out[j] = tile[row j][col L&15].  Lane addresses are
+// independent, so the tile rows may live at ANY stride — a 16x16x32 MFMA
+// B-fragment can be read straight from a ROW-MAJOR LDS image (two reads:
+// k = 8g..8g+3 and 8g+4..8g+7, g = lane>>4), eliminating the separate
+// transposed staging image and its 8 scalar LDS writes per thread.
+// Chunk->lane semantics hardware-verified (tools: tr16_probe).
+template <int DT> struct TrRead;
+template <> struct TrRead<2> {
+  typedef __attribute__((ext_vector_type(4))) __bf16 v4;
+  static __device__ __forceinline__ void rd(const unsigned short* p, void* out) {
+    v4 r = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (__attribute__((address_space(3))) v4*)(uintptr_t)(const void*)p);
+    __builtin_memcpy(out, &r, 8);
+  }
+};
+template <> struct TrRead<1> {
+  typedef __attribute__((ext_vector_type(4))) __fp16 v4;
+  static __device__ __forceinline__ void rd(const _Float16* p, void* out) {
+    v4 r = __builtin_amdgcn_ds_read_tr16_b64_v4f16(
+        (__attribute__((address_space(3))) v4*)(uintptr_t)(const void*)p);
+    __builtin_memcpy(out, &r, 8);
+  }
+};
+// B-fragment (16x16x32 map: lane l -> col l&15, k = 8*(l>>4)+j) from a
+// row-major image img[row][ds]: rows row0.., cols col0..col0+15.
+template <int DT>
+__device__ __forceinline__ typename MFMA16<DT>::frag trread_bfrag(
+    const typename DTraits<DT>::T* img, int row0, int col0, int ds, int lane) {
+  const int rb = row0 + 8 * (lane >> 4) + ((lane & 15) >> 2);
+  const int cb = col0 + 4 * (lane & 3);
+  typename MFMA16<DT>::frag f;
+  TrRead<DT>::rd(img + rb * ds + cb, &f);
+  TrRead<DT>::rd(img + (rb + 4) * ds + cb, (char*)&f + 8);
+  return f;
+}
+
 #define NEG_BIG (-1e30f)
 
 // row-group shuffle reduce: combine over the 16 lanes that share l>>4
@@ -557,8 +595,9 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
 // grid over (b, hq, kv-tile of 64 keys); wave owns 16 keys.  Loop q tiles of
 // 32 with double-buffered async staging (v2, same structure as forward) and
 // XOR-swizzled transposed images (xmask 3: 64-B data span per row).
-// LDS: Q[2][32][D+8] | Q_T[2][D][32+8] | dO[2][32][D+8] | dO_T[2][D][32+8] |
-//      lse[2][32] f32 | delta[2][32] f32 | P_T[4][16][32+8] | dS[4][16][32+8]
+// LDS: Q[2][32][D+8] | dO[2][32][D+8] | lse[2][32] f32 | delta[2][32] f32 |
+//      P_T[8][16][32+8] | dS[8][16][32+8]   (dV/dK B-frags come straight
+//      from the row-major Q/dO images via ds_read_b64_tr_b16)
 template <int DT, int D>
 __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     typename DTraits<DT>::T* __restrict__ dk_out,
@@ -584,10 +623,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   T* Q_lds = (T*)smem_raw;                   // [2][QT][DS]
-  T* QT_lds = Q_lds + 2 * QT * DS;           // [2][D][QS] swizzled
-  T* dO_lds = QT_lds + 2 * D * QS;           // [2][QT][DS]
-  T* dOT_lds = dO_lds + 2 * QT * DS;         // [2][D][QS] swizzled
-  T* PT_lds = dOT_lds + 2 * D * QS;          // [8][16][QS]  (P^T tiles)
+  T* dO_lds = Q_lds + 2 * QT * DS;           // [2][QT][DS]
+  T* PT_lds = dO_lds + 2 * QT * DS;          // [8][16][QS]  (P^T tiles)
   T* DS_lds = PT_lds + 8 * 16 * QS;          // [8][16][QS]  (dS^T tiles; separate
                                              //  buffer: avoids an LDS WAR hazard
                                              //  between the P^T A-frag read and
@@ -660,13 +697,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     if (st_on) {
       *(frag*)(Q_lds + buf * QT * DS + st_row * DS + st_c8) = qreg;
       *(frag*)(dO_lds + buf * QT * DS + st_row * DS + st_c8) = dreg;
-      T* QTb = QT_lds + buf * D * QS;
-      T* dOTb = dOT_lds + buf * D * QS;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        *vt_addr(QTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&qreg)[j];
-        *vt_addr(dOTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&dreg)[j];
-      }
     }
     if (st_t < QT) {
       lse_lds[buf * QT + st_t] = lse_reg;
@@ -683,8 +713,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const int cur = (qt - qstart) & 1;
     T* Qb = Q_lds + cur * QT * DS;
     T* dOb = dO_lds + cur * QT * DS;
-    T* QTb = QT_lds + cur * D * QS;
-    T* dOTb = dOT_lds + cur * D * QS;
     const float* lse_b = lse_lds + cur * QT;
     const float* dl_b = dl_lds + cur * QT;
     if (qt + 1 < nQT2) load_qtile(qt + 1);
@@ -734,7 +762,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     frag pa = *(const frag*)(Pw + lo * QS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
-      frag bd = *(const frag*)vt_addr(dOTb, dn * 16 + lo, hi * 8 * 2, QS * 2, 3);
+      frag bd = trread_bfrag<DT>(dOb, 0, dn * 16, DS, lane);
       dv_acc[dn] = MF::mma(pa, bd, dv_acc[dn]);
     }
 
@@ -747,7 +775,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     frag da = *(const frag*)(Dw + lo * QS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
-      frag bq = *(const frag*)vt_addr(QTb, dn * 16 + lo, hi * 8 * 2, QS * 2, 3);
+      frag bq = trread_bfrag<DT>(Qb, 0, dn * 16, DS, lane);
       dk_acc[dn] = MF::mma(da, bq, dk_acc[dn]);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -799,8 +827,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   T* K_lds = (T*)smem_raw;                    // [2][KT][DS]
   T* V_lds = K_lds + 2 * KT * DS;             // [2][KT][DS]
-  T* KT_lds = V_lds + 2 * KT * DS;            // [2][D][KS] swizzled (xmask 3)
-  T* S_lds = KT_lds + 2 * D * KS;             // [8][16][KS]
+  T* S_lds = V_lds + 2 * KT * DS;             // [8][16][KS]  (dQ B-frags come
+                                              //  from row-major K via tr_read)
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -867,10 +895,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     if (st_on) {
       *(frag*)(K_lds + buf * KT * DS + st_row * DS + st_c8) = kreg;
       *(frag*)(V_lds + buf * KT * DS + st_row * DS + st_c8) = vreg;
-      T* KTb = KT_lds + buf * D * KS;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *vt_addr(KTb, st_c8 + j, st_row * 2, KS * 2, 3) = ((const T*)&kreg)[j];
     }
   };
 
@@ -883,7 +907,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const int cur = kt & 1;
     T* Kb = K_lds + cur * KT * DS;
     T* Vb = V_lds + cur * KT * DS;
-    T* KTb = KT_lds + cur * D * KS;
     if (kt + 1 < n_kt) load_ktile(kt + 1);
 
     // wave-uniform diagonal skip (8-wave skew): if every key in this kv
@@ -921,7 +944,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     frag da = *(const frag*)(Sw + lo * KS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
-      frag bk = *(const frag*)vt_addr(KTb, dn * 16 + lo, hi * 8 * 2, KS * 2, 3);
+      frag bk = trread_bfrag<DT>(Kb, 0, dn * 16, DS, lane);
       dq_acc[dn] = MF::mma(da, bk, dq_acc[dn]);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -1442,7 +1465,7 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
   }
   const int nKT = (int)((S + 127) / 128);    // 8-wave WG: 128 keys
   const int grid = (int)(B * Hq * nKT);
-  const size_t lds = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 8 * 16 * QS) + sizeof(float) * 4 * QT;
+  const size_t lds = sizeof(T) * (4 * QT * DS + 2 * 8 * 16 * QS) + sizeof(float) * 4 * QT;
   hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
                      (const T*)q, (const T*)k, (const T*)v, lse, delta,
@@ -1494,7 +1517,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   }
   const int nQT = (int)((S + 127) / 128);    // 8-wave WG: 128 q rows
   const int grid = (int)(B * Hq * nQT);
-  const size_t lds = sizeof(T) * (4 * KT * DS + 2 * D * KS + 8 * 16 * KS);
+  const size_t lds = sizeof(T) * (4 * KT * DS + 8 * 16 * KS);
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
