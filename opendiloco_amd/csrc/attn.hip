@@ -145,8 +145,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   T* K_lds = (T*)smem_raw;                       // [2][KT][DS]
-  T* VT_lds = K_lds + 2 * KT * DS;               // [2][D][KS] swizzled
-  T* P_lds = VT_lds + 2 * D * KS;                // [4][16][KS]
+  T* V_lds = K_lds + 2 * KT * DS;                // [2][KT][DS] row-major
+  T* P_lds = V_lds + 2 * KT * DS;                // [4][16][KS]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -202,13 +202,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   };
   auto write_tile = [&](int buf) {
     T* Kb = K_lds + buf * KT * DS;
-    T* Vb = VT_lds + buf * D * KS;
+    T* Vb = V_lds + buf * KT * DS;
 #pragma unroll
     for (int i = 0; i < LPT; ++i) {
       *(frag*)(Kb + st_row[i] * DS + st_c8[i]) = kreg[i];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *vt_addr(Vb, st_c8[i] + j, st_row[i] * 2, KS * 2) = ((const T*)&vreg[i])[j];
+      *(frag*)(Vb + st_row[i] * DS + st_c8[i]) = vreg[i];
     }
   };
 
@@ -220,7 +218,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const int kbase = kt * KT;
     const int cur = kt & 1;
     T* Kb = K_lds + cur * KT * DS;
-    T* Vb = VT_lds + cur * D * KS;
+    T* Vb = V_lds + cur * KT * DS;
     if (kt + 1 < n_kt) load_tile(kt + 1);  // async: in flight during compute
 
     // ---- S tile = Q K^T (16 q x KT keys) ----
@@ -294,7 +292,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     for (int dn = 0; dn < NDN; ++dn)
 #pragma unroll
       for (int c = 0; c < KT / 32; ++c) {
-        frag bv = *(const frag*)vt_addr(Vb, dn * 16 + lo, (c * 32 + hi * 8) * 2, KS * 2);
+        frag bv = trread_bfrag<DT>(Vb, c * 32, dn * 16, DS, lane);
         o_acc[dn] = MF::mma(pa[c], bv, o_acc[dn]);
       }
     __builtin_amdgcn_s_setprio(0);
@@ -344,8 +342,24 @@ template <> struct MFMA32<1> {
   }
 };
 
+// A-fragment for 32x32x16 MFMA (map: lane l -> row l&31, k = 8*(l>>5)+j)
+// transposed out of a ROW-MAJOR image img[k][ds] via ds_read_b64_tr_b16:
+// out[j] = img[row0 + 8*(l>>5) + j][col0 + (l&31)].  The four 16-lane tr
+// groups split as (k-half = l>>5) x (col-half = (l>>4)&1).
+template <int DT>
+__device__ __forceinline__ typename MFMA32<DT>::frag trread_afrag32(
+    const typename DTraits<DT>::T* img, int row0, int col0, int ds, int lane) {
+  const int lam = lane & 15;
+  const int rb = row0 + 8 * (lane >> 5) + (lam >> 2);
+  const int cb = col0 + 16 * ((lane >> 4) & 1) + 4 * (lam & 3);
+  typename MFMA32<DT>::frag f;
+  TrRead<DT>::rd(img + rb * ds + cb, &f);
+  TrRead<DT>::rd(img + (rb + 4) * ds + cb, (char*)&f + 8);
+  return f;
+}
+
 template <int DT, int D>
-__global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
+__global__ __launch_bounds__(256, 4) void attn_fwd_v3_kernel(
     typename DTraits<DT>::T* __restrict__ o, float* __restrict__ lse,
     const typename DTraits<DT>::T* __restrict__ q,
     const typename DTraits<DT>::T* __restrict__ k,
@@ -366,7 +380,7 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   T* K_lds = (T*)smem_raw;                       // [2][KT][DS]
-  T* VT_lds = K_lds + 2 * KT * DS;               // [2][D][KS] swizzled
+  T* V_lds = K_lds + 2 * KT * DS;                // [2][KT][DS] row-major
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -420,13 +434,11 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
   };
   auto write_tile = [&](int buf) {
     T* Kb = K_lds + buf * KT * DS;
-    T* Vb = VT_lds + buf * D * KS;
+    T* Vb = V_lds + buf * KT * DS;
 #pragma unroll
     for (int i = 0; i < LPT; ++i) {
       *(shortx8*)(Kb + st_row[i] * DS + st_c8[i]) = kreg[i];
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *vt_addr(Vb, st_c8[i] + j, st_row[i] * 2, KS * 2) = ((const T*)&vreg[i])[j];
+      *(shortx8*)(Vb + st_row[i] * DS + st_c8[i]) = vreg[i];
     }
   };
 
@@ -438,7 +450,7 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
     const int kbase = kt * KT;
     const int cur = kt & 1;
     T* Kb = K_lds + cur * KT * DS;
-    T* Vb = VT_lds + cur * D * KS;
+    T* Vb = V_lds + cur * KT * DS;
     if (kt + 1 < n_kt) load_tile(kt + 1);
 
 #pragma unroll
@@ -526,8 +538,8 @@ __global__ __launch_bounds__(256, 3) void attn_fwd_v3_kernel(
       for (int mt = 0; mt < NMT; ++mt)
 #pragma unroll
         for (int c = 0; c < 2; ++c) {
-          frag va = *(const frag*)vt_addr(Vb, mt * 32 + lo32,
-                                          (st * 32 + c * 16 + hi5 * 8) * 2, KS * 2);
+          frag va = trread_afrag32<DT>(Vb, st * 32 + c * 16, mt * 32, DS,
+                                       (int)(threadIdx.x & 63));
           oacc[mt] = MF::mma(va, pfrag[c], oacc[mt]);
         }
       __builtin_amdgcn_s_setprio(0);
@@ -1382,7 +1394,7 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
     // v3: swapped QK^T, in-register softmax, 128 q rows per workgroup
     const int nQT = (int)((S + 127) / 128);
     const int grid = (int)(B * Hq * nQT);
-    const size_t lds = sizeof(T) * (2 * KT * DS + 2 * D * KS);
+    const size_t lds = sizeof(T) * (2 * KT * DS + 2 * KT * DS);
     hipLaunchKernelGGL((attn_fwd_v3_kernel<DT, D>), dim3(grid), dim3(256), lds,
                        (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
                        (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale,
@@ -1392,7 +1404,7 @@ static int launch_attn_fwd(void* o, float* lse, const void* q, const void* k,
   }
   const int nQT = (int)((S + 63) / 64);
   const int grid = (int)(B * Hq * nQT);
-  const size_t lds = sizeof(T) * (2 * KT * DS + 2 * D * KS + 4 * 16 * KS);
+  const size_t lds = sizeof(T) * (2 * KT * DS + 2 * KT * DS + 4 * 16 * KS);
   hipLaunchKernelGGL((attn_fwd_kernel<DT, D>), dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (T*)o, lse, (const T*)q, (const T*)k,
                      (const T*)v, (int)B, (int)Hq, (int)Hkv, (int)S, scale,
