@@ -1006,10 +1006,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   T* Q_lds = (T*)smem_raw;                   // [2][QT][DS]
-  T* QT_lds = Q_lds + 2 * QT * DS;           // [2][D][QS] swizzled (xmask 3)
-  T* dO_lds = QT_lds + 2 * D * QS;           // [2][QT][DS]
-  T* dOT_lds = dO_lds + 2 * QT * DS;         // [2][D][QS] swizzled
-  T* PT_lds = dOT_lds + 2 * D * QS;          // [4][32][QS]
+  T* dO_lds = Q_lds + 2 * QT * DS;           // [2][QT][DS]
+  T* PT_lds = dO_lds + 2 * QT * DS;          // [4][32][QS]
   T* DST_lds = PT_lds + 4 * 32 * QS;         // [4][32][QS]
   float* lse_lds = (float*)(DST_lds + 4 * 32 * QS);  // [2][QT]
   float* dl_lds = lse_lds + 2 * QT;                  // [2][QT]
@@ -1071,13 +1069,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
     if (st_on) {
       *(shortx8*)(Q_lds + buf * QT * DS + st_row * DS + st_c8) = qreg;
       *(shortx8*)(dO_lds + buf * QT * DS + st_row * DS + st_c8) = dreg;
-      T* QTb = QT_lds + buf * D * QS;
-      T* dOTb = dOT_lds + buf * D * QS;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        *vt_addr(QTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&qreg)[j];
-        *vt_addr(dOTb, st_c8 + j, st_row * 2, QS * 2, 3) = ((const T*)&dreg)[j];
-      }
     }
     if (st_t < QT) {
       lse_lds[buf * QT + st_t] = lse_reg;
@@ -1094,8 +1085,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
     const int cur = (qt2 - qstart) & 1;
     T* Qb = Q_lds + cur * QT * DS;
     T* dOb = dO_lds + cur * QT * DS;
-    T* QTb = QT_lds + cur * D * QS;
-    T* dOTb = dOT_lds + cur * D * QS;
     const float* lse_b = lse_lds + cur * QT;
     const float* dl_b = dl_lds + cur * QT;
     if (qt2 + 1 < nQT2) load_qtile(qt2 + 1);
@@ -1148,7 +1137,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
     for (int mt = 0; mt < NMT; ++mt)
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        frag bd = *(const frag*)vt_addr(dOTb, mt * 32 + lo32, (c * 16 + hi5 * 8) * 2, QS * 2, 3);
+        frag bd = trread_afrag32<DT>(dOb, c * 16, mt * 32, DS, lane);
         dv_acc[mt] = MF::mma(pa[c], bd, dv_acc[mt]);
       }
     __builtin_amdgcn_s_setprio(0);
@@ -1167,7 +1156,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
     for (int mt = 0; mt < NMT; ++mt)
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        frag bq = *(const frag*)vt_addr(QTb, mt * 32 + lo32, (c * 16 + hi5 * 8) * 2, QS * 2, 3);
+        frag bq = trread_afrag32<DT>(Qb, c * 16, mt * 32, DS, lane);
         dk_acc[mt] = MF::mma(da[c], bq, dk_acc[mt]);
       }
     __builtin_amdgcn_s_setprio(0);
@@ -1212,8 +1201,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_v3_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   T* K_lds = (T*)smem_raw;                    // [2][KT][DS]
   T* V_lds = K_lds + 2 * KT * DS;             // [2][KT][DS]
-  T* KT_lds = V_lds + 2 * KT * DS;            // [2][D][KS] swizzled (xmask 3)
-  T* S_lds = KT_lds + 2 * D * KS;             // [4][32][KS]
+  T* S_lds = V_lds + 2 * KT * DS;             // [4][32][KS]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -1278,10 +1266,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_v3_kernel(
     if (st_on) {
       *(shortx8*)(K_lds + buf * KT * DS + st_row * DS + st_c8) = kreg;
       *(shortx8*)(V_lds + buf * KT * DS + st_row * DS + st_c8) = vreg;
-      T* KTb = KT_lds + buf * D * KS;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *vt_addr(KTb, st_c8 + j, st_row * 2, KS * 2, 3) = ((const T*)&kreg)[j];
     }
   };
 
@@ -1294,7 +1278,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_v3_kernel(
     const int cur = kt2 & 1;
     T* Kb = K_lds + cur * KT * DS;
     T* Vb = V_lds + cur * KT * DS;
-    T* KTb = KT_lds + cur * D * KS;
     if (kt2 + 1 < n_kt) load_ktile(kt2 + 1);
 
     // ---- S = Q K^T and dP = dO V^T  (C: col = key = lane&31) ----
@@ -1340,7 +1323,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_v3_kernel(
     for (int mt = 0; mt < NMT; ++mt)
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        frag bk = *(const frag*)vt_addr(KTb, mt * 32 + lo32, (c * 16 + hi5 * 8) * 2, KS * 2, 3);
+        frag bk = trread_afrag32<DT>(Kb, c * 16, mt * 32, DS, lane);
         dq_acc[mt] = MF::mma(da[c], bk, dq_acc[mt]);
       }
     __builtin_amdgcn_s_setprio(0);
@@ -1466,7 +1449,7 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
   if (use_bwd_v3() && v_sb == 0 && dv_sb == 0) {  // v3 port is contiguous-only
     const int nKT3 = (int)((S + 127) / 128);
     const int grid3 = (int)(B * Hq * nKT3);
-    const size_t lds3 = sizeof(T) * (4 * QT * DS + 4 * D * QS + 2 * 4 * 32 * QS)
+    const size_t lds3 = sizeof(T) * (4 * QT * DS + 2 * 4 * 32 * QS)
                         + sizeof(float) * 4 * QT;
     hipLaunchKernelGGL((attn_bwd_dkdv_v3_kernel<DT, D>), dim3(grid3), dim3(256), lds3,
                        (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
@@ -1519,7 +1502,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   if (use_bwd_v3() && v_sb == 0) {  // v3 port is contiguous-only
     const int nQT3 = (int)((S + 127) / 128);
     const int grid3 = (int)(B * Hq * nQT3);
-    const size_t lds3 = sizeof(T) * (4 * KT * DS + 2 * D * KS + 4 * 32 * KS);
+    const size_t lds3 = sizeof(T) * (4 * KT * DS + 4 * 32 * KS);
     hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<DT, D>), dim3(grid3), dim3(256), lds3,
                        (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                        (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
