@@ -497,37 +497,40 @@ __global__ void rope_move_kernel(typename DTraits<DT>::T* __restrict__ out,
   using TR = DTraits<DT>;
   using T = typename TR::T;
   const int hd = D / 2;
-  const int nq = hd / 4;  // 4-pair quads per row
+  const int nq = hd / 8;  // 8-pair pieces per row (b128 loads both halves)
   const int64_t total = total_rows * nq;
   for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
     const int64_t row = idx / nq;
-    const int i0 = (int)(idx % nq) * 4;
+    const int i0 = (int)(idx % nq) * 8;
     const int sp = (int)(row % S);
     const int hh = (int)((row / S) % H);
     const int64_t b = row / ((int64_t)S * H);
     const T* irow = in + b * i_sb + hh * i_sh + sp * i_sr;
     T* orow = out + b * o_sb + hh * o_sh + sp * o_sr;
-    shortx4 x1 = *(const shortx4*)(irow + i0);
-    shortx4 x2 = *(const shortx4*)(irow + i0 + hd);
+    shortx8 x1 = *(const shortx8*)(irow + i0);
+    shortx8 x2 = *(const shortx8*)(irow + i0 + hd);
     if (ROT) {
-      const float4 cv = *(const float4*)(costab + (int64_t)sp * hd + i0);
-      const float4 sv = *(const float4*)(sintab + (int64_t)sp * hd + i0);
-      shortx4 o1, o2;
+      const float4 cv0 = *(const float4*)(costab + (int64_t)sp * hd + i0);
+      const float4 cv1 = *(const float4*)(costab + (int64_t)sp * hd + i0 + 4);
+      const float4 sv0 = *(const float4*)(sintab + (int64_t)sp * hd + i0);
+      const float4 sv1 = *(const float4*)(sintab + (int64_t)sp * hd + i0 + 4);
+      shortx8 o1, o2;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < 8; ++j) {
         float a = TR::toF(((const T*)&x1)[j]);
         float bb = TR::toF(((const T*)&x2)[j]);
-        float c = ((const float*)&cv)[j];
-        float sn = BWD ? -((const float*)&sv)[j] : ((const float*)&sv)[j];
+        float c = j < 4 ? ((const float*)&cv0)[j] : ((const float*)&cv1)[j - 4];
+        float sn = j < 4 ? ((const float*)&sv0)[j] : ((const float*)&sv1)[j - 4];
+        if (BWD) sn = -sn;
         ((T*)&o1)[j] = TR::fromF(a * c - bb * sn);
         ((T*)&o2)[j] = TR::fromF(bb * c + a * sn);
       }
-      *(shortx4*)(orow + i0) = o1;
-      *(shortx4*)(orow + i0 + hd) = o2;
+      *(shortx8*)(orow + i0) = o1;
+      *(shortx8*)(orow + i0 + hd) = o2;
     } else {
-      *(shortx4*)(orow + i0) = x1;
-      *(shortx4*)(orow + i0 + hd) = x2;
+      *(shortx8*)(orow + i0) = x1;
+      *(shortx8*)(orow + i0 + hd) = x2;
     }
   }
 }
@@ -537,9 +540,9 @@ extern "C" int dk_rope_move(void* out, const void* in, const float* costab,
                             int64_t D, int64_t i_sb, int64_t i_sh, int64_t i_sr,
                             int64_t o_sb, int64_t o_sh, int64_t o_sr, int backward,
                             int rotate, int dtype, dkStream stream) {
-  if ((D / 2) % 4 != 0 || dtype == 0) return (int)hipErrorInvalidValue;
+  if ((D / 2) % 8 != 0 || dtype == 0) return (int)hipErrorInvalidValue;
   const int64_t total_rows = B * H * S;
-  int grid = dk_stream_grid(total_rows * (D / 2) / 4, 256);
+  int grid = dk_stream_grid(total_rows * (D / 2) / 8, 256);
   DK_DISPATCH_DT(dtype, {
     if constexpr (kDT != 0) {
       using T = typename DTraits<kDT>::T;
