@@ -830,7 +830,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   using T = typename TR::T;
   using MF = MFMA16<DT>;
   using frag = typename MF::frag;
-  constexpr int KT = 32;
+  constexpr int KT = 64;                      // kv tile (halves barrier count)
   constexpr int KS = KT + 8;
   constexpr int DS = D + 8;
   constexpr int NKC = D / 32;
@@ -924,10 +924,10 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     // wave-uniform diagonal skip (8-wave skew): if every key in this kv
     // tile exceeds this wave's last q row, the whole tile is masked to zero.
     if (kbase < q0 + 16) {
-    float ds[2][4];
+    float ds[4][4];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
+    for (int nt = 0; nt < 4; ++nt) {
       floatx4 st = (floatx4)(0.f), dpt = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
@@ -947,18 +947,23 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       }
     }
 
-    // dQ += dS K  (A = dS via LDS; B = K_T)
+    // dQ += dS K  (A = dS via LDS; B from row-major K via tr_read);
+    // KT=64 keys = two 32-deep contraction chunks
     T* Sw = S_lds + wave * 16 * KS;
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt)
+    for (int nt = 0; nt < 4; ++nt)
 #pragma unroll
       for (int r = 0; r < 4; ++r) Sw[(hi * 4 + r) * KS + nt * 16 + lo] = TR::fromF(ds[nt][r]);
-    frag da = *(const frag*)(Sw + lo * KS + hi * 8);
+    frag da[2];
 #pragma unroll
-    for (int dn = 0; dn < NDN; ++dn) {
-      frag bk = trread_bfrag<DT>(Kb, 0, dn * 16, DS, lane);
-      dq_acc[dn] = MF::mma(da, bk, dq_acc[dn]);
-    }
+    for (int c = 0; c < 2; ++c) da[c] = *(const frag*)(Sw + lo * KS + c * 32 + hi * 8);
+#pragma unroll
+    for (int dn = 0; dn < NDN; ++dn)
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        frag bk = trread_bfrag<DT>(Kb, c * 32, dn * 16, DS, lane);
+        dq_acc[dn] = MF::mma(da[c], bk, dq_acc[dn]);
+      }
     __builtin_amdgcn_s_setprio(0);
     }  // end diagonal skip
 
@@ -1498,7 +1503,8 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
                               int64_t v_sb, int64_t v_sh, int64_t v_sr,
                               dkStream stream) {
   using T = typename DTraits<DT>::T;
-  constexpr int KT = 32, KS = KT + 8, DS = D + 8;
+  constexpr int KT = 32, KS = KT + 8, DS = D + 8;   // v3 tile constants
+  constexpr int KT2 = 64, KS2 = KT2 + 8;            // v2 8-wave tile constants
   if (use_bwd_v3() && v_sb == 0) {  // v3 port is contiguous-only
     const int nQT3 = (int)((S + 127) / 128);
     const int grid3 = (int)(B * Hq * nQT3);
@@ -1512,7 +1518,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   }
   const int nQT = (int)((S + 127) / 128);    // 8-wave WG: 128 q rows
   const int grid = (int)(B * Hq * nQT);
-  const size_t lds = sizeof(T) * (4 * KT * DS + 8 * 16 * KS);
+  const size_t lds = sizeof(T) * (4 * KT2 * DS + 8 * 16 * KS2);
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
