@@ -104,23 +104,10 @@ __device__ __forceinline__ float grp16_sum(float x) {
 // v2 structure (guide T14/T3/T5): KT=64 kv tiles, double-buffered LDS, async
 // register staging (next tile's global loads issued before compute, written
 // to the other LDS buffer after it — HBM latency hides under MFMA), ONE
-// barrier per tile, XOR-swizzled transposed V image (kills the staging
-// write bank conflicts), s_setprio(1) around the MFMA clusters.
-// LDS: K[2][KT][DS] | V_T[2][D][KS (swizzled)] | P[4][16][KS]
-
-// byte-level XOR swizzle for the transposed-V image: row d's 16-B groups are
-// rotated by (d>>3)&7 — spreads the 8-u16-scatter staging writes over 8
-// banks and keeps every 16-B read aligned (XOR bits 4-6 only; row data span
-// 128 B < 144 B stride so the XOR never leaves the row).
-// xmask limits the XOR to 16-B groups that stay inside the row's data span:
-// 7 for a 128-B span (KS=72 u16 rows), 3 for a 64-B span (QS=40 u16 rows).
-template <typename T>
-__device__ __forceinline__ T* vt_addr(T* base, int d, int kbyte_off, int KSbytes,
-                                      int xmask = 7) {
-  int byte = d * KSbytes + kbyte_off;
-  byte ^= ((d >> 3) & xmask) << 4;
-  return (T*)((char*)base + byte);
-}
+// barrier per tile, both K and V staged ROW-MAJOR (PV fragments are
+// transposed at read time by ds_read_b64_tr_b16), s_setprio(1) around the
+// MFMA clusters.
+// LDS: K[2][KT][DS] | V[2][KT][DS] | P[4][16][KS]
 
 template <int DT, int D>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
@@ -372,7 +359,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_v3_kernel(
   using MF = MFMA32<DT>;
   using frag = typename MF::frag;
   constexpr int KT = 64;            // kv tile (2 x 32-key subtiles)
-  constexpr int KS = KT + 8;        // V^T image stride (u16), XOR-swizzled
+  constexpr int KS = KT + 8;        // P image stride (u16)
   constexpr int DS = D + 8;         // K image stride
   constexpr int NKC = D / 16;       // 16-channel contraction chunks for QK^T
   constexpr int NMT = D / 32;       // 32-row d tiles for O^T
@@ -532,7 +519,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_v3_kernel(
         pfrag[c] = *(frag*)&pw;
       }
 
-      // ---- O^T += V^T P  (A = V^T from the swizzled image, B = P) ----
+      // ---- O^T += V^T P  (A = V^T tr_read from row-major V, B = P) ----
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mt = 0; mt < NMT; ++mt)
@@ -604,9 +591,9 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
 }
 
 // ======================= bwd dK/dV =======================
-// grid over (b, hq, kv-tile of 64 keys); wave owns 16 keys.  Loop q tiles of
-// 32 with double-buffered async staging (v2, same structure as forward) and
-// XOR-swizzled transposed images (xmask 3: 64-B data span per row).
+// grid over (b, hq, kv-tile of 128 keys); 8 waves x 16 keys each.  Loop q
+// tiles of 32 with double-buffered async staging (v2 structure); dV/dK
+// B-fragments tr_read straight from the row-major Q/dO images.
 // LDS: Q[2][32][D+8] | dO[2][32][D+8] | lse[2][32] f32 | delta[2][32] f32 |
 //      P_T[8][16][32+8] | dS[8][16][32+8]   (dV/dK B-frags come straight
 //      from the row-major Q/dO images via ds_read_b64_tr_b16)
@@ -813,8 +800,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 }
 
 // ======================= bwd dQ =======================
-// grid over (b, hq, q-tile of 64); wave owns 16 q rows.  Loop kv tiles of 32.
-// LDS: K[32][D+8] | V[32][D+8] | K_T[D][32+8] | dS[4][16][32+8]
+// grid over (b, hq, q-tile of 128); 8 waves x 16 q rows each.  Loop kv
+// tiles of 64; dQ B-fragments tr_read from the row-major K image.
+// LDS: K[2][64][D+8] | V[2][64][D+8] | dS[8][16][64+8]
 template <int DT, int D>
 __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     typename DTraits<DT>::T* __restrict__ dq_out,
