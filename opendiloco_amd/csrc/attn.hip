@@ -614,22 +614,23 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   using T = typename TR::T;
   using MF = MFMA16<DT>;
   using frag = typename MF::frag;
-  constexpr int QT = 32;
+  constexpr int QT = 32;                     // compute half-tile (q rows)
+  constexpr int QTT = 64;                    // staged tile: 2 halves / barrier
   constexpr int QS = QT + 8;
   constexpr int DS = D + 8;
   constexpr int NKC = D / 32;
   constexpr int NDN = D / 16;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  T* Q_lds = (T*)smem_raw;                   // [2][QT][DS]
-  T* dO_lds = Q_lds + 2 * QT * DS;           // [2][QT][DS]
-  T* PT_lds = dO_lds + 2 * QT * DS;          // [8][16][QS]  (P^T tiles)
+  T* Q_lds = (T*)smem_raw;                   // [2][QTT][DS]
+  T* dO_lds = Q_lds + 2 * QTT * DS;          // [2][QTT][DS]
+  T* PT_lds = dO_lds + 2 * QTT * DS;         // [8][16][QS]  (P^T tiles)
   T* DS_lds = PT_lds + 8 * 16 * QS;          // [8][16][QS]  (dS^T tiles; separate
                                              //  buffer: avoids an LDS WAR hazard
                                              //  between the P^T A-frag read and
                                              //  the dS^T writes in one iteration)
-  float* lse_lds = (float*)(DS_lds + 8 * 16 * QS);  // [2][QT]
-  float* dl_lds = lse_lds + 2 * QT;                 // [2][QT]
+  float* lse_lds = (float*)(DS_lds + 8 * 16 * QS);  // [2][QTT]
+  float* dl_lds = lse_lds + 2 * QTT;                // [2][QTT]
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -666,12 +667,12 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 #pragma unroll
   for (int dn = 0; dn < NDN; ++dn) { dv_acc[dn] = (floatx4)(0.f); dk_acc[dn] = (floatx4)(0.f); }
 
-  const int qstart = (kt * 128) / QT;        // first q tile that sees these keys
-  const int nQT2 = (S + QT - 1) / QT;
+  const int qstart = (kt * 128) / QTT;       // first staged tile with these keys
+  const int nQT2 = (S + QTT - 1) / QTT;
 
-  // async staging state (QT*D/8 <= 256 loads: one piece per thread)
+  // async staging state (QTT*D/8 = 512 loads: one piece per thread)
   const int st_t = (int)threadIdx.x;
-  const bool st_on = st_t < (QT * D) / 8;
+  const bool st_on = st_t < (QTT * D) / 8;
   const int st_row = st_t / (D / 8);
   const int st_c8 = (st_t % (D / 8)) * 8;
   frag qreg, dreg;
@@ -679,14 +680,14 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 
   const int64_t gbase = (int64_t)b * g_sb + (int64_t)h * g_sh;
   auto load_qtile = [&](int qt) {
-    const int qrow = qt * QT + st_row;
+    const int qrow = qt * QTT + st_row;
     const int qr_c = qrow < S ? qrow : S - 1;
     if (st_on) {
       qreg = *(const frag*)(q + qoff + (int64_t)qr_c * D + st_c8);
       dreg = *(const frag*)(do_ + gbase + (int64_t)qr_c * g_sr + st_c8);
     }
-    if (st_t < QT) {
-      const int rr = qt * QT + st_t;
+    if (st_t < QTT) {
+      const int rr = qt * QTT + st_t;
       const int rr_c = rr < S ? rr : S - 1;
       lse_reg = lse[lseoff + rr_c];
       dl_reg = delta[lseoff + rr_c];
@@ -694,12 +695,12 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   };
   auto write_qtile = [&](int buf) {
     if (st_on) {
-      *(frag*)(Q_lds + buf * QT * DS + st_row * DS + st_c8) = qreg;
-      *(frag*)(dO_lds + buf * QT * DS + st_row * DS + st_c8) = dreg;
+      *(frag*)(Q_lds + buf * QTT * DS + st_row * DS + st_c8) = qreg;
+      *(frag*)(dO_lds + buf * QTT * DS + st_row * DS + st_c8) = dreg;
     }
-    if (st_t < QT) {
-      lse_lds[buf * QT + st_t] = lse_reg;
-      dl_lds[buf * QT + st_t] = dl_reg;
+    if (st_t < QTT) {
+      lse_lds[buf * QTT + st_t] = lse_reg;
+      dl_lds[buf * QTT + st_t] = dl_reg;
     }
   };
 
@@ -708,17 +709,23 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   __syncthreads();
 
   for (int qt = qstart; qt < nQT2; ++qt) {
-    const int qbase = qt * QT;
     const int cur = (qt - qstart) & 1;
-    T* Qb = Q_lds + cur * QT * DS;
-    T* dOb = dO_lds + cur * QT * DS;
-    const float* lse_b = lse_lds + cur * QT;
-    const float* dl_b = dl_lds + cur * QT;
+    T* Q64 = Q_lds + cur * QTT * DS;
+    T* dO64 = dO_lds + cur * QTT * DS;
+    const float* lse64 = lse_lds + cur * QTT;
+    const float* dl64 = dl_lds + cur * QTT;
     if (qt + 1 < nQT2) load_qtile(qt + 1);
 
+    // two 32-q compute halves per staged 64-row tile: one barrier per 64 q
+#pragma unroll
+    for (int hf = 0; hf < 2; ++hf) {
+    const int qbase = qt * QTT + hf * QT;
+    T* Qb = Q64 + hf * QT * DS;
+    T* dOb = dO64 + hf * QT * DS;
+    const float* lse_b = lse64 + hf * QT;
+    const float* dl_b = dl64 + hf * QT;
     // 8-wave WGs skew the diagonal: a wave whose 16 keys all sit above this
-    // q tile (k0 > every qcol) would compute an all-masked (zero) tile —
-    // skip the MFMAs (wave-uniform branch); staging + barrier still run.
+    // q half (k0 > every qcol) computes an all-masked (zero) tile — skip.
     if (qbase + QT > k0) {
     // ---- S^T = K Q^T (16 keys x 32 q), P^T = exp(scale*S^T - lse) ----
     float pt[2][4], dst[2][4];
@@ -779,6 +786,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
     }  // end diagonal skip
+    }  // end half loop
 
     if (qt + 1 < nQT2) write_qtile(cur ^ 1);  // T14: write late
     __syncthreads();
@@ -1453,7 +1461,8 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
   }
   const int nKT = (int)((S + 127) / 128);    // 8-wave WG: 128 keys
   const int grid = (int)(B * Hq * nKT);
-  const size_t lds = sizeof(T) * (4 * QT * DS + 2 * 8 * 16 * QS) + sizeof(float) * 4 * QT;
+  constexpr int QTT = 64;  // staged q tile (2 compute halves per barrier)
+  const size_t lds = sizeof(T) * (4 * QTT * DS + 2 * 8 * 16 * QS) + sizeof(float) * 4 * QTT;
   hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dk_o, (T*)dv_o, (const T*)do_,
                      (const T*)q, (const T*)k, (const T*)v, lse, delta,
