@@ -115,14 +115,26 @@ def test_rejects_reference_forbidden_kwargs():
 
 # ---------- 2-process gloo: the cross-worker average ----------
 
-def _worker(rank, world, port, outdir):
+def _worker(rank, world, port, outdir, compression=None):
     os.environ.update(dict(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
                            MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port)))
     import torch.distributed as dist
 
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
-        model, opt = _make_opt(seed=0, H=2)  # same init on both ranks
+        if compression is None:
+            model, opt = _make_opt(seed=0, H=2)  # same init on both ranks
+        else:
+            torch.manual_seed(0)
+            model = torch.nn.Linear(5, 1)
+            opt = DiLoCoOptimizer(
+                dht=None, run_id="test", batch_size=32, num_inner_steps=2,
+                outer_optimizer=partial(torch.optim.SGD, lr=0.7, momentum=0.9,
+                                        nesterov=True),
+                inner_optimizer=partial(torch.optim.AdamW, lr=0.1, weight_decay=0.1,
+                                        betas=(0.9, 0.95)),
+                params=model.parameters(), grad_compression=compression)
+            assert opt.diloco_grad_averager.comm_dtype == torch.float16
         _train_steps(model, opt, 2, seed=50 + rank)  # different data per rank
         # after the outer round every worker must hold identical params
         torch.save({"flat": opt.flat.flat_param.clone(), "epoch": opt.local_epoch},
@@ -148,3 +160,27 @@ def test_two_worker_average_converges_to_same_params(tmp_path):
     r1 = torch.load(tmp_path / "out_1.pt", weights_only=False)
     assert r0["epoch"] == 1 and r1["epoch"] == 1
     assert torch.allclose(r0["flat"], r1["flat"], atol=1e-7)
+
+
+def test_two_worker_fp16_compression(tmp_path):
+    """`--hv.hivemind_compression fp16` twin (reference Float16Compression,
+    utils.py:83-121): the pseudo-gradient all-reduce payload is fp16; both
+    workers must still land on identical finite params."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, str(tmp_path), "fp16"))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(180)
+        assert p.exitcode == 0
+    r0 = torch.load(tmp_path / "out_0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "out_1.pt", weights_only=False)
+    assert r0["epoch"] == 1 and r1["epoch"] == 1
+    assert torch.equal(r0["flat"], r1["flat"])
+    assert torch.isfinite(r0["flat"]).all()
