@@ -43,5 +43,7 @@ def test_torch_binding_importable(repo_root):
 
     mod = load_binding()
     for fn in ["rmsnorm_fwd", "rope", "swiglu_fwd", "ce_fwd", "attn_fwd", "attn_bwd",
-               "fused_adamw", "clip_grad_", "pseudo_grad", "outer_nesterov", "probe_mfma"]:
+               "attn_fwd_bsd", "attn_bwd_bsd", "qkv_rope_gather", "rope_scatter_",
+               "accum_", "fused_adamw", "clip_grad_", "pseudo_grad", "outer_nesterov",
+               "probe_mfma"]:
         assert hasattr(mod, fn), fn
