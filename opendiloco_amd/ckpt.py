@@ -61,13 +61,21 @@ def get_resume_info(ckpt_config) -> tuple[bool, str | None]:
 
 def save_checkpoint(checkpoint_path: str, model, optimizer, scheduler,
                     outer_optimizer=None, scaler=None, loss: float | None = None,
-                    data_loader=None, save_global_state: bool = True, rank: int = 0):
-    """ckpt_utils.py:48-100 analogue (see module docstring for the format)."""
+                    data_loader=None, save_global_state: bool = True,
+                    save_model_optim: bool = True, rank: int = 0):
+    """ckpt_utils.py:48-100 analogue (see module docstring for the format).
+
+    ``save_model_optim=False`` lets non-zero ranks of a replicated (non-hv)
+    world skip the model_optim.pt write: the states are identical across
+    ranks after the grad all-reduce, and concurrent writes of the same file
+    could interleave/corrupt it.
+    """
     os.makedirs(checkpoint_path, exist_ok=True)
-    torch.save({
-        "model": model.state_dict(),
-        "optimizer": optimizer.state_dict(),
-    }, os.path.join(checkpoint_path, MODEL_OPTIM_FILE))
+    if save_model_optim:
+        torch.save({
+            "model": model.state_dict(),
+            "optimizer": optimizer.state_dict(),
+        }, os.path.join(checkpoint_path, MODEL_OPTIM_FILE))
     if data_loader is not None:
         torch.save({"data_loader": data_loader.state_dict()},
                    os.path.join(checkpoint_path, f"__{rank}_0.pt"))

@@ -149,7 +149,13 @@ class DiLoCoGradAverager:
             # the consuming outer kernel is restored via stream waits
             self._comm_stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self._comm_stream):
-                dist.all_reduce(send, op=dist.ReduceOp.AVG)
+                if backend == "gloo":
+                    # gloo has no ReduceOp.AVG (reachable via the library API
+                    # with CUDA tensors on gloo); SUM + divide is the same mean
+                    dist.all_reduce(send, op=dist.ReduceOp.SUM)
+                    send.div_(ws)
+                else:
+                    dist.all_reduce(send, op=dist.ReduceOp.AVG)
             torch.cuda.current_stream().wait_stream(self._comm_stream)
         else:
             if backend == "gloo":

@@ -277,7 +277,20 @@ class _CrossEntropyFn(torch.autograd.Function):
 
 
 def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-    """Shifted mean CE over full [B, S, V] logits vs [B, S] labels."""
+    """Shifted mean CE over full [B, S, V] logits vs [B, S] labels.
+
+    No ignore_index: the training path feeds labels = input_ids (fake-data
+    collation, reference utils.py:163-167), never HF's -100 padding.  A
+    negative label would be an out-of-bounds gather index in the GPU kernel,
+    so it is rejected here (cheap: min over int64 [B, S] on device, checked
+    lazily with the loss; the sync happens at .item() on the loss anyway).
+    """
+    if int(labels.min()) < 0:
+        raise ValueError(
+            "causal_lm_loss does not support ignore_index labels (< 0); "
+            "got labels.min() < 0. Mask/pad handling is out of scope for "
+            "the fake-data training path (see DESIGN.md)."
+        )
     return _CrossEntropyFn.apply(logits, labels)
 
 

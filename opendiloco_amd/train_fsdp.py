@@ -261,14 +261,13 @@ def train(config: Config):
             # H steps, train_fsdp.py:410-413; a worker is exactly 1 GPU here)
 
             if rank == 0:
-                galaxy_size = config.hv.galaxy_size if hv else 1
-                if hv:
-                    galaxy_size = max(galaxy_size, world_size)
                 total_samples = real_step * config.total_batch_size
                 effective_step = real_step
                 if hv:
-                    effective_step = real_step * galaxy_size
-                    total_samples = real_step * config.total_batch_size * galaxy_size
+                    # reference uses config.hv.galaxy_size as-is
+                    # (train_fsdp.py:418-423; "not robust to off/on ramping")
+                    effective_step = real_step * config.hv.galaxy_size
+                    total_samples = real_step * config.total_batch_size * config.hv.galaxy_size
                 metrics = {
                     "Loss": loss_batch.item(),
                     "step": real_step,
@@ -309,11 +308,17 @@ def train(config: Config):
                             loss=loss_batch.item(), scaler=scaler,
                             data_loader=train_dataloader, save_global_state=True, rank=rank)
                 else:
+                    # model/optimizer states are replicated across ranks after
+                    # the grad all-reduce: only rank 0 writes model_optim.pt
+                    # (concurrent same-path writes could corrupt it); every
+                    # rank still writes its own dataloader state file
                     save_checkpoint(
                         checkpoint_path=ckpt_path, model=model, optimizer=optimizer,
                         scheduler=scheduler, loss=loss_batch.item(), scaler=scaler,
                         data_loader=train_dataloader, save_global_state=(rank == 0),
-                        rank=rank)
+                        save_model_optim=(rank == 0), rank=rank)
+                    if world_size > 1:
+                        dist.barrier()
                 if local_rank == 0 and config.ckpt.topk is not None:
                     deleted = delete_old_checkpoints(config.ckpt.path, config.ckpt.topk)
                     if deleted:
