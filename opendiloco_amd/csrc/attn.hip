@@ -86,6 +86,11 @@ __device__ __forceinline__ typename MFMA16<DT>::frag trread_bfrag(
   return f;
 }
 
+// 4-wide 16-bit vector store (one ds_write_b64) for the packed P/dS images
+template <int DT> struct Pack4;
+template <> struct Pack4<2> { using V = shortx4; };
+template <> struct Pack4<1> { typedef __attribute__((ext_vector_type(4))) _Float16 V; };
+
 #define NEG_BIG (-1e30f)
 
 // row-group shuffle reduce: combine over the 16 lanes that share l>>4
@@ -759,26 +764,41 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
       }
     }
 
-    // ---- dV += P^T dO  (A = P^T via LDS; B = dO_T) ----
-    T* Pw = PT_lds + wave * 16 * QS;
+    // ---- dV += P^T dO  (A = P^T; B = dO_T) ----
+    // P staged ROW-MAJOR [32 q][16 keys] per wave with ONE packed
+    // ds_write_b64 per lane per 16-q block (each lane owns 4 consecutive
+    // keys of one q column) instead of 16 scalar ds_write_b16; the
+    // A-fragment of P^T then comes straight out of the row-major P image
+    // via ds_read_b64_tr_b16 (A and B fragment lane maps coincide).
+    // PS = 16 + 4 pad: bank stride 10 dwords -> the 16 packed writes of a
+    // lane group land on 16 distinct banks.
+    constexpr int PS = 20;
+    T* Pw = PT_lds + wave * 16 * QS;   // reuse the per-wave region (640 elems)
+    using P4 = typename Pack4<DT>::V;
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt)
+    for (int nt = 0; nt < 2; ++nt) {
+      P4 pk;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) Pw[(hi * 4 + r) * QS + nt * 16 + lo] = TR::fromF(pt[nt][r]);
-    frag pa = *(const frag*)(Pw + lo * QS + hi * 8);
+      for (int r = 0; r < 4; ++r) ((T*)&pk)[r] = TR::fromF(pt[nt][r]);
+      *(P4*)(Pw + (nt * 16 + lo) * PS + hi * 4) = pk;
+    }
+    frag pa = trread_bfrag<DT>(Pw, 0, 0, PS, lane);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
       frag bd = trread_bfrag<DT>(dOb, 0, dn * 16, DS, lane);
       dv_acc[dn] = MF::mma(pa, bd, dv_acc[dn]);
     }
 
-    // ---- dK += dS^T Q  (A = dS^T via LDS; B = Q_T) ----
+    // ---- dK += dS^T Q  (A = dS^T via the same packed row-major image) ----
     T* Dw = DS_lds + wave * 16 * QS;
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt)
+    for (int nt = 0; nt < 2; ++nt) {
+      P4 dk4;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) Dw[(hi * 4 + r) * QS + nt * 16 + lo] = TR::fromF(dst[nt][r]);
-    frag da = *(const frag*)(Dw + lo * QS + hi * 8);
+      for (int r = 0; r < 4; ++r) ((T*)&dk4)[r] = TR::fromF(dst[nt][r]);
+      *(P4*)(Dw + (nt * 16 + lo) * PS + hi * 4) = dk4;
+    }
+    frag da = trread_bfrag<DT>(Dw, 0, 0, PS, lane);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
       frag bq = trread_bfrag<DT>(Qb, 0, dn * 16, DS, lane);
@@ -943,16 +963,25 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       }
     }
 
-    // dQ += dS K  (A = dS via LDS; B from row-major K via tr_read);
-    // KT=64 keys = two 32-deep contraction chunks
-    T* Sw = S_lds + wave * 16 * KS;
+    // dQ += dS K  (A = dS; B from row-major K via tr_read); KT=64 keys =
+    // two 32-deep contraction chunks.  dS staged ROW-MAJOR [64 key][16 q]
+    // per wave with ONE packed ds_write_b64 per lane per 16-key block
+    // (lane owns 4 consecutive q rows of one key column) instead of 16
+    // scalar ds_write_b16; the A-fragments of dS then come straight out of
+    // the image via ds_read_b64_tr_b16 (PS as in dkdv).
+    constexpr int PS = 20;
+    T* Sw = S_lds + wave * 64 * PS;
+    using P4 = typename Pack4<DT>::V;
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt)
+    for (int nt = 0; nt < 4; ++nt) {
+      P4 s4;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) Sw[(hi * 4 + r) * KS + nt * 16 + lo] = TR::fromF(ds[nt][r]);
+      for (int r = 0; r < 4; ++r) ((T*)&s4)[r] = TR::fromF(ds[nt][r]);
+      *(P4*)(Sw + (nt * 16 + lo) * PS + hi * 4) = s4;
+    }
     frag da[2];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) da[c] = *(const frag*)(Sw + lo * KS + c * 32 + hi * 8);
+    for (int c = 0; c < 2; ++c) da[c] = trread_bfrag<DT>(Sw, c * 32, 0, PS, lane);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn)
 #pragma unroll
@@ -1515,7 +1544,8 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   }
   const int nQT = (int)((S + 127) / 128);    // 8-wave WG: 128 q rows
   const int grid = (int)(B * Hq * nQT);
-  const size_t lds = sizeof(T) * (4 * KT2 * DS + 8 * 16 * KS2);
+  (void)KS2;
+  const size_t lds = sizeof(T) * (4 * KT2 * DS + 8 * 64 * 20);  // dS [64][PS=20]/wave
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
