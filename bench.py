@@ -59,6 +59,12 @@ def main():
     ap.add_argument("--precision", default="bf16", choices=["bf16", "fp16"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--cpu-baseline-only", action="store_true")
+    ap.add_argument("--advance-steps", type=int, default=0,
+                    help="pre-advance the DiLoCo progress tracker by N inner steps "
+                         "(no compute) so a short timed window can cross an outer "
+                         "boundary at large H; e.g. --h 500 --advance-steps 497 "
+                         "--warmup 2 --steps 6 times steps 500-505 incl. the outer "
+                         "round at 500 (the H=500 sustained leg)")
     args = ap.parse_args()
 
     dist = _dist_init()
@@ -116,6 +122,13 @@ def main():
         clip_grad_norm_flat_(opt.flat.flat_grad, 1.0)
         opt.step()
         opt.zero_grad()
+
+    if args.advance_steps:
+        # tracker-only advance: the inner steps are identical regardless of
+        # the sample counter; this lets the timed region cross the H-step
+        # outer boundary without paying H-3 untimed real steps first
+        opt.tracker.report_local_progress(
+            opt.local_epoch, samples_accumulated=args.advance_steps * args.batch)
 
     for _ in range(args.warmup):
         one_step()
@@ -191,6 +204,9 @@ def main():
             "traffic": traffic,
             "kernel": "dk_fused_adamw",
             "launch_ms": dur_s * 1000.0,
+            # MFMA-class fractions (attention + GEMM classes, HIP-event timed
+            # at the bench shape) beside the AdamW HBM story
+            "mfma": _mfma_fractions(args, mcfg),
         }
 
     cpu_baseline = None
@@ -235,20 +251,89 @@ def main():
         dist.destroy_process_group()
 
 
+def _mfma_fractions(args, mcfg) -> dict:
+    """Attention fwd/bwd and GEMM-class achieved TFLOP/s vs the measured
+    dense bf16 MFMA ceiling (2495 TF, MI355X_MICROARCH.md 32x32x16 row),
+    HIP-event timed at the bench shape.  Read with DESIGN.md §3: the
+    attention kernels run on 16x16x32 tiles, whose per-SIMD issue ceiling is
+    ~half the 32x32 rate."""
+    import torch.nn.functional as F
+
+    from opendiloco_amd.ops import _ext
+
+    ext = _ext()
+    dtype = torch.bfloat16 if args.precision == "bf16" else torch.float16
+    B, Hq, S = args.per_device, mcfg.num_attention_heads, args.seq
+    D = mcfg.hidden_size // mcfg.num_attention_heads
+    PEAK_TF = 2495.0
+
+    def timeit(fn, reps=10):
+        e0, e1 = torch.cuda.Event(enable_timing=True), torch.cuda.Event(enable_timing=True)
+        for _ in range(3):
+            fn()
+        torch.cuda.synchronize()
+        e0.record()
+        for _ in range(reps):
+            fn()
+        e1.record()
+        torch.cuda.synchronize()
+        return e0.elapsed_time(e1) / 1000.0 / reps
+
+    out = {"peak_tflops": PEAK_TF, "peak_note": "measured dense bf16 32x32x16 ceiling"}
+    q = torch.randn(B, Hq, S, D, device="cuda", dtype=dtype)
+    k, v = torch.randn_like(q), torch.randn_like(q)
+    scale = D ** -0.5
+    attn_flops = 4.0 * B * Hq * S * S * D / 2  # causal
+    dt = timeit(lambda: ext.attn_fwd(q, k, v, scale))
+    out["attn_fwd"] = {"tflops": attn_flops / dt / 1e12, "frac": attn_flops / dt / 1e12 / PEAK_TF}
+    o, lse = ext.attn_fwd(q, k, v, scale)
+    do = torch.randn_like(o)
+    dt = timeit(lambda: ext.attn_bwd(do, q, k, v, o, lse, scale))
+    out["attn_bwd"] = {"tflops": 2.5 * attn_flops / dt / 1e12,
+                       "frac": 2.5 * attn_flops / dt / 1e12 / PEAK_TF}
+    del q, k, v, o, lse, do
+    # GEMM classes (library kernels) over the model's projection shapes
+    R = args.per_device * args.seq
+    h = mcfg.hidden_size
+    shapes = [(3 * h, h, mcfg.num_hidden_layers), (h, h, mcfg.num_hidden_layers),
+              (2 * mcfg.intermediate_size, h, mcfg.num_hidden_layers),
+              (h, mcfg.intermediate_size, mcfg.num_hidden_layers),
+              (mcfg.vocab_size, h, 1)]
+    tot_t = {"fwd": 0.0, "dx": 0.0, "dw": 0.0}
+    tot_f = 0.0
+    for out_f, in_f, n in shapes:
+        x = torch.randn(R, in_f, device="cuda", dtype=dtype)
+        w = torch.randn(out_f, in_f, device="cuda", dtype=dtype)
+        dy = torch.randn(R, out_f, device="cuda", dtype=dtype)
+        flops = 2.0 * R * out_f * in_f
+        tot_f += flops * n
+        tot_t["fwd"] += timeit(lambda: F.linear(x, w)) * n
+        tot_t["dx"] += timeit(lambda: dy @ w) * n
+        tot_t["dw"] += timeit(lambda: dy.t() @ x) * n
+        del x, w, dy
+    for kind in tot_t:
+        tf = tot_f / tot_t[kind] / 1e12
+        out[f"gemm_{kind}"] = {"tflops": tf, "frac": tf / PEAK_TF}
+    return out
+
+
 def _cpu_baseline(args, mcfg) -> dict:
-    """Reference CPU path (the oracle restatement of train_diloco_torch.py,
-    transformers fp32 on host cores) on a BOUNDED sample: one
-    fwd+bwd+clip+AdamW micro-step of 1 sequence, extrapolated to tokens/s."""
+    """Reference CPU path timed on host cores: the FULL BASELINE.json
+    configs[0] loop — llama-150m(-shaped), 1 worker, H=1, per-device-batch 8,
+    ONE real step = 1 micro fwd+bwd (8x1024 tokens) + clip + AdamW +
+    scheduler + the outer block (pseudo-grad, 1-worker mean, Nesterov SGD,
+    re-offload), restated from train_diloco_torch.py:272-353 with the
+    oracle's own pieces (kind "port")."""
     import torch
 
-    from oracle.diloco_oracle import OracleConfig, make_reference_model
+    from oracle.diloco_oracle import (OracleConfig, _fake_batch,
+                                      get_cosine_schedule_with_warmup,
+                                      make_reference_model)
 
-    cfg_path = os.path.join(REPO_ROOT, "tests", "models", "llama-2m")  # placeholder path
-    ocfg = OracleConfig(model_path=cfg_path, fresh_init_seed=42, seq_length=args.seq,
-                        vocab_size=mcfg.vocab_size)
-    # build the reference model at the bench shape
     import tempfile
 
+    ocfg = OracleConfig(model_path="", fresh_init_seed=42, seq_length=args.seq,
+                        vocab_size=mcfg.vocab_size)
     with tempfile.TemporaryDirectory() as d:
         with open(os.path.join(d, "config.json"), "w") as f:
             json.dump({"architectures": ["LlamaForCausalLM"], "model_type": "llama",
@@ -262,27 +347,40 @@ def _cpu_baseline(args, mcfg) -> dict:
                        "vocab_size": mcfg.vocab_size}, f)
         ocfg.model_path = d
         model = make_reference_model(ocfg).train()
+    bs = 8  # configs[0]: per-device-batch 8, grad_acc 1, H = 1
     inner = torch.optim.AdamW(model.parameters(), lr=4e-4, weight_decay=0.1, betas=(0.9, 0.95))
+    outer = torch.optim.SGD(model.parameters(), lr=0.7, momentum=0.9, nesterov=True)
+    sched = get_cosine_schedule_with_warmup(inner, 1000, 88_000)
+    offloaded = [p.data.detach().clone() for g in outer.param_groups for p in g["params"]]
     gen = torch.Generator().manual_seed(42)
-    n_seq = 1
-    ids = torch.randint(3, mcfg.vocab_size, (n_seq, args.seq), generator=gen)
-    batch = dict(input_ids=ids, attention_mask=torch.ones_like(ids), labels=ids.clone())
-    # one untimed warm step
+    batch = _fake_batch(gen, bs, args.seq, mcfg.vocab_size)
+    # one untimed warm micro-step (allocator/MKL warmup)
     model(**batch).loss.backward()
     inner.zero_grad()
+    batch = _fake_batch(gen, bs, args.seq, mcfg.vocab_size)
     t0 = time.perf_counter()
-    loss = model(**batch).loss
-    loss.backward()
-    torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
-    inner.step()
-    inner.zero_grad()
+    loss = model(**batch).loss  # train_diloco_torch.py:313
+    loss.backward()  # :318
+    torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)  # :323
+    inner.step()  # :325
+    sched.step()  # :327
+    inner.zero_grad()  # :334
+    # outer block (H=1: fires every real step), :336-353
+    main_param = [p for g in inner.param_groups for p in g["params"]]
+    for off, p in zip(offloaded, main_param):
+        p.grad = off.data - p.data  # 1-worker all_reduce(AVG) = identity
+        p.data = off.data.clone()
+    outer.step()
+    outer.zero_grad()
+    offloaded = [p.data.detach().clone() for g in outer.param_groups for p in g["params"]]
     dt = time.perf_counter() - t0
     return {
-        "value": n_seq * args.seq / dt,
+        "value": bs * args.seq / dt,
         "unit": "tokens/s",
         "cores": torch.get_num_threads(),
         "kind": "port",
-        "sample": f"1 inner micro-step ({n_seq}x{args.seq} tokens fwd+bwd+clip+AdamW), "
+        "sample": f"full configs[0] real step ({bs}x{args.seq} tokens fwd+bwd+clip+AdamW"
+                  f"+scheduler + outer pseudo-grad/Nesterov/re-offload at H=1), "
                   f"transformers fp32 on host cores ({dt:.1f}s)",
     }
 

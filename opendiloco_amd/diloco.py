@@ -116,7 +116,7 @@ class DiLoCoGradAverager:
                  comm_dtype: torch.dtype = torch.float32):
         self.flat = flat
         self.flat_outer = flat_outer
-        self.comm_dtype = comm_dtype
+        self.comm_dtype = comm_dtype  # float32 | float16 | uint8 (uniform8bit)
         self.pseudo_grad = torch.zeros_like(flat_outer)
         self._comm_stream = (torch.cuda.Stream() if flat_outer.device.type == "cuda" else None)
         self.last_allreduce_seconds: float = 0.0
@@ -133,11 +133,47 @@ class DiLoCoGradAverager:
         self.compute_and_load_pseudo_grad_into_averager()
         self.all_reduce_()
 
+    # uniform 8-bit payload codec: restates the published algorithm of
+    # hivemind's Uniform8BitQuantization (hivemind @ 213bff9,
+    # requirements.txt:7 — source NOT vendored under the reference, so this
+    # is a restatement, not a copy; reference call site utils.py:103-107):
+    # values are quantized to 256 uniform levels spanning RANGE_IN_SIGMAS
+    # standard deviations around the mean; each peer's payload is
+    # de-quantized before averaging (the averager decompresses peers'
+    # tensors and means them in fp32).  The reference pins this codec only
+    # as "non-NaN averaged pseudo-grads" (test_diloco_hivemind.py:90-93);
+    # parity proper is anchored at the uncompressed torch-only twin
+    # (SURVEY.md §8c).
+    RANGE_IN_SIGMAS = 6.0
+
+    def _uniform8bit_allreduce(self, buf: torch.Tensor, ws: int) -> None:
+        offset = buf.mean()
+        scale = (self.RANGE_IN_SIGMAS * buf.std() / 255.0).clamp_min(torch.finfo(torch.float32).tiny)
+        q = torch.clamp(torch.round((buf - offset) / scale) + 128.0, 0.0, 255.0).to(torch.uint8)
+        meta = torch.stack([offset, scale])
+        q_all = [torch.empty_like(q) for _ in range(ws)]
+        meta_all = [torch.empty_like(meta) for _ in range(ws)]
+        dist.all_gather(q_all, q)
+        dist.all_gather(meta_all, meta)
+        buf.zero_()
+        for qi, mi in zip(q_all, meta_all):
+            buf.add_(qi.to(torch.float32).sub_(128.0).mul_(mi[1]).add_(mi[0]))
+        buf.div_(ws)
+
     def all_reduce_(self) -> None:
         ws = _world_size()
         if ws <= 1:
             return
         buf = self.pseudo_grad
+        if self.comm_dtype == torch.uint8:
+            if self._comm_stream is not None:
+                self._comm_stream.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(self._comm_stream):
+                    self._uniform8bit_allreduce(buf, ws)
+                torch.cuda.current_stream().wait_stream(self._comm_stream)
+            else:
+                self._uniform8bit_allreduce(buf, ws)
+            return
         reduced_in_lower_precision = self.comm_dtype != torch.float32
         if reduced_in_lower_precision:
             send = buf.to(self.comm_dtype)
@@ -260,8 +296,11 @@ class DiLoCoOptimizer:
         name = type(grad_compression).__name__ if grad_compression is not None else ""
         if isinstance(grad_compression, str):
             name = grad_compression
-        if "fp16" in name.lower() or "float16" in name.lower():
+        lname = name.lower()
+        if "fp16" in lname or "float16" in lname:
             comm_dtype = torch.float16
+        elif "uniform8bit" in lname:
+            comm_dtype = torch.uint8
         self.diloco_grad_averager = DiLoCoGradAverager(self.flat, self.state_averager.flat_outer,
                                                        comm_dtype=comm_dtype)
 

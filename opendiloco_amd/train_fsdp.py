@@ -250,7 +250,24 @@ def train(config: Config):
                 torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
 
             if hv:
-                optimizer.step(scaler=scaler)
+                try:
+                    optimizer.step(scaler=scaler)
+                except RuntimeError as e:
+                    # a dead peer surfaces as a failed collective on the fixed
+                    # RCCL/gloo world; map it to the reference's rank-drop
+                    # error shape (train_fsdp.py:452-457).  The reference can
+                    # keep training without the lost peer (elastic DHT swarm);
+                    # a fixed communicator cannot — fail fast either way,
+                    # with the reference's message when fail_rank_drop is set.
+                    num_peers = max(optimizer.tracker.global_progress.num_peers - 1, 0)
+                    optimizer.tracker.global_progress.num_peers = num_peers
+                    log(f"Lost a diloco worker, num_peers: {num_peers}, "
+                        f"galaxy_size: {config.hv.galaxy_size}")
+                    if config.hv.fail_rank_drop:
+                        raise ValueError(
+                            f"Lost a diloco worker, num_peers: {num_peers}, "
+                            f"galaxy_size: {config.hv.galaxy_size}") from e
+                    raise
             else:
                 scaler.step(optimizer)
             scaler.update()
@@ -285,7 +302,10 @@ def train(config: Config):
                     max_num_peers = max(max_num_peers, num_peers)
                     metrics["num_peers"] = num_peers
                     if num_peers < max_num_peers and config.hv.fail_rank_drop:
-                        raise ValueError(f"Lost a diloco worker, num_peers: {num_peers}")
+                        # reference train_fsdp.py:455-457 message shape
+                        raise ValueError(
+                            f"Lost a diloco worker, num_peers: {num_peers}, "
+                            f"galaxy_size: {config.hv.galaxy_size}")
                 if logging_activations_steps:
                     metrics.update({k: float(v) for k, v in log_activations.items()})
                     log_activations = {}
@@ -325,6 +345,16 @@ def train(config: Config):
                         log(f"Deleted old checkpoints: {deleted}")
 
             loss_batch = torch.zeros((), device=device)
+
+            # test-only fault injection for the straggler harness
+            # (tests/test_train_cli.py::test_rank_drop_fails_fast): simulate a
+            # worker dying without a clean process-group shutdown
+            die = os.environ.get("DILOCO_TEST_DIE_RANK_STEP")
+            if die:
+                die_rank, die_step = (int(x) for x in die.split(":"))
+                if rank == die_rank and real_step >= die_step:
+                    log(f"test fault injection: rank {rank} exiting at step {real_step}")
+                    os._exit(0)
 
             if config.max_steps is not None and real_step >= config.max_steps:
                 break

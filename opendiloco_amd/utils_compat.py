@@ -84,10 +84,15 @@ def get_compression_kwargs(hivemind_compression: str | None) -> dict:
         return {"grad_compression": None}
     if hivemind_compression in ("fp16", "scaled-fp16"):
         return {"grad_compression": "fp16"}
-    if hivemind_compression in ("uniform8bit", "quantile8bit", "blockwise8bit"):
+    if hivemind_compression == "uniform8bit":
+        # uniform 256-level quantization of the flat payload (restated from
+        # hivemind's Uniform8BitQuantization; see DiLoCoGradAverager)
+        return {"grad_compression": "uniform8bit"}
+    if hivemind_compression in ("quantile8bit", "blockwise8bit"):
         raise NotImplementedError(
-            f"hivemind_compression={hivemind_compression}: 8-bit codecs are not "
-            f"implemented in the RCCL backend (use fp16 or none)")
+            f"hivemind_compression={hivemind_compression}: only the uniform8bit "
+            f"codec is implemented on the RCCL flat payload (use uniform8bit, "
+            f"fp16 or none) — permanent single-node deviation, DESIGN.md §6")
     raise ValueError(f"Invalid hivemind_compression: {hivemind_compression}")
 
 

