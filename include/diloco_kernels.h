@@ -181,6 +181,14 @@ int dk_outer_nesterov(float* theta_outer, float* theta_local, float* momentum_bu
  * train_diloco_torch.py:305-310 grad flow). */
 int dk_accum(float* dst, const void* src, int64_t n, int src_dtype,
              dkStream stream);
+
+/* fp32 dst[i] += sum_b src[b*chunk_stride + i] — deterministic reduction of
+ * the split-K weight-gradient partial slabs into the fp32 master gradient
+ * (the dW = dy^T x GEMMs of every projection, reference F.linear backward
+ * inside transformers Llama at train_fsdp.py:383; split over the token
+ * dimension to fill all 256 CUs). */
+int dk_accum_chunks(float* dst, const float* src, int64_t n, int nchunk,
+                    int64_t chunk_stride, dkStream stream);
 int dk_cast(void* dst, const void* src, int64_t n, int dst_dtype, int src_dtype,
             dkStream stream);
 

@@ -732,73 +732,69 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     // 8-wave WGs skew the diagonal: a wave whose 16 keys all sit above this
     // q half (k0 > every qcol) computes an all-masked (zero) tile — skip.
     if (qbase + QT > k0) {
-    // ---- S^T = K Q^T (16 keys x 32 q), P^T = exp(scale*S^T - lse) ----
+    // ---- S = Q K^T computed TRANSPOSED-C: mma(Q_as_A, K_as_B) ----
+    // The 16x16x32 A and B fragments share one lane map (row/col = l&15,
+    // k = 8*(l>>4)+j), so swapping the operands flips the C orientation
+    // for free: C row = q (hi*4+r), col = key (lo).  Each lane then holds
+    // 4 CONSECUTIVE q values for one key, which pack into the P^T/dS^T
+    // images ([16 key][QS] row-major, exactly the round-1 layout) as ONE
+    // ds_write_b64 instead of 16 scalar ds_write_b16 per image; the
+    // A-fragment reads stay the round-1 contiguous b128 reads, and the
+    // per-q lse/delta lookups become one float4 read per 16-q block.
     float pt[2][4], dst[2][4];
     __builtin_amdgcn_s_setprio(1);
+    const int krow = k0 + lo;
 #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
       floatx4 st = (floatx4)(0.f);
-#pragma unroll
-      for (int kc = 0; kc < NKC; ++kc) {
-        frag bq = *(const frag*)(Qb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
-        st = MF::mma(k_frag[kc], bq, st);
-      }
-      // dP^T = V dO^T
       floatx4 dpt = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
-        frag bd = *(const frag*)(dOb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
-        dpt = MF::mma(v_frag[kc], bd, dpt);
+        frag qa = *(const frag*)(Qb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        st = MF::mma(qa, k_frag[kc], st);
+        frag doa = *(const frag*)(dOb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        dpt = MF::mma(doa, v_frag[kc], dpt);
       }
-      const int qcol = qbase + nt * 16 + lo;
-      const float lse_q = lse_b[nt * 16 + lo];
-      const float dl_q = dl_b[nt * 16 + lo];
+      const floatx4 lse4 = *(const floatx4*)(lse_b + nt * 16 + hi * 4);
+      const floatx4 dl4 = *(const floatx4*)(dl_b + nt * 16 + hi * 4);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int krow = k0 + hi * 4 + r;
+        const int qrow = qbase + nt * 16 + hi * 4 + r;
         float pv = 0.f;
-        if (krow <= qcol && krow < S && qcol < S)
-          pv = __expf(st[r] * scale - lse_q);
+        if (krow <= qrow && krow < S && qrow < S)
+          pv = __expf(st[r] * scale - lse4[r]);
         pt[nt][r] = pv;
-        dst[nt][r] = pv * (dpt[r] - dl_q) * scale;
+        dst[nt][r] = pv * (dpt[r] - dl4[r]) * scale;
       }
     }
 
-    // ---- dV += P^T dO  (A = P^T; B = dO_T) ----
-    // P staged ROW-MAJOR [32 q][16 keys] per wave with ONE packed
-    // ds_write_b64 per lane per 16-q block (each lane owns 4 consecutive
-    // keys of one q column) instead of 16 scalar ds_write_b16; the
-    // A-fragment of P^T then comes straight out of the row-major P image
-    // via ds_read_b64_tr_b16 (A and B fragment lane maps coincide).
-    // PS = 16 + 4 pad: bank stride 10 dwords -> the 16 packed writes of a
-    // lane group land on 16 distinct banks.
-    constexpr int PS = 20;
-    T* Pw = PT_lds + wave * 16 * QS;   // reuse the per-wave region (640 elems)
+    // ---- dV += P^T dO  (A = P^T via packed per-wave LDS image; B = dO_T) ----
+    T* Pw = PT_lds + wave * 16 * QS;
     using P4 = typename Pack4<DT>::V;
 #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
       P4 pk;
 #pragma unroll
       for (int r = 0; r < 4; ++r) ((T*)&pk)[r] = TR::fromF(pt[nt][r]);
-      *(P4*)(Pw + (nt * 16 + lo) * PS + hi * 4) = pk;
+      *(P4*)(Pw + lo * QS + nt * 16 + hi * 4) = pk;
     }
-    frag pa = trread_bfrag<DT>(Pw, 0, 0, PS, lane);
+    frag pa = *(const frag*)(Pw + lo * QS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
       frag bd = trread_bfrag<DT>(dOb, 0, dn * 16, DS, lane);
       dv_acc[dn] = MF::mma(pa, bd, dv_acc[dn]);
     }
 
-    // ---- dK += dS^T Q  (A = dS^T via the same packed row-major image) ----
+    // ---- dK += dS^T Q  (A = dS^T via the same packed image layout) ----
     T* Dw = DS_lds + wave * 16 * QS;
 #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
       P4 dk4;
 #pragma unroll
       for (int r = 0; r < 4; ++r) ((T*)&dk4)[r] = TR::fromF(dst[nt][r]);
-      *(P4*)(Dw + (nt * 16 + lo) * PS + hi * 4) = dk4;
+      *(P4*)(Dw + lo * QS + nt * 16 + hi * 4) = dk4;
     }
-    frag da = trread_bfrag<DT>(Dw, 0, 0, PS, lane);
+    frag da = *(const frag*)(Dw + lo * QS + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn) {
       frag bq = trread_bfrag<DT>(Qb, 0, dn * 16, DS, lane);
@@ -878,23 +874,18 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   const int64_t v_rs = v_sb ? v_sr : (int64_t)D;
 
   frag q_frag[NKC], do_frag[NKC];
-  float lse_r[4], dl_r[4];
+  float lse_q, dl_q;
+  const int qrow_l = q0 + lo;   // this lane's q row (C col after the operand swap)
   {
-    const int qrow = q0 + lo;
-    const int qr_c = qrow < S ? qrow : S - 1;
+    const int qr_c = qrow_l < S ? qrow_l : S - 1;
     const int64_t gbase = (int64_t)b * g_sb + (int64_t)h * g_sh;
 #pragma unroll
     for (int kc = 0; kc < NKC; ++kc) {
       q_frag[kc] = *(const frag*)(q + qoff + (int64_t)qr_c * D + kc * 32 + hi * 8);
       do_frag[kc] = *(const frag*)(do_ + gbase + (int64_t)qr_c * g_sr + kc * 32 + hi * 8);
     }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int rr = q0 + hi * 4 + r;
-      const int rr_c = rr < S ? rr : S - 1;
-      lse_r[r] = lse[lseoff + rr_c];
-      dl_r[r] = delta[lseoff + rr_c];
-    }
+    lse_q = lse[lseoff + qr_c];
+    dl_q = delta[lseoff + qr_c];
   }
 
   floatx4 dq_acc[NDN];
@@ -940,6 +931,12 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     // wave-uniform diagonal skip (8-wave skew): if every key in this kv
     // tile exceeds this wave's last q row, the whole tile is masked to zero.
     if (kbase < q0 + 16) {
+    // S computed TRANSPOSED-C (see the dkdv kernel): mma(K_as_A, Q_as_B)
+    // puts C row = key (hi*4+r), col = q (lo), so each lane holds 4
+    // CONSECUTIVE keys of one q row — the dS^T image ([16 q][KS] row-major,
+    // round-1 layout) takes ONE packed ds_write_b64 per 16-key block
+    // instead of 16 scalar ds_write_b16, and the A-fragment reads stay the
+    // round-1 contiguous b128 reads.
     float ds[4][4];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -947,41 +944,35 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       floatx4 st = (floatx4)(0.f), dpt = (floatx4)(0.f);
 #pragma unroll
       for (int kc = 0; kc < NKC; ++kc) {
-        frag bk = *(const frag*)(Kb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
-        st = MF::mma(q_frag[kc], bk, st);
-        frag bv = *(const frag*)(Vb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
-        dpt = MF::mma(do_frag[kc], bv, dpt);
+        frag ka = *(const frag*)(Kb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        st = MF::mma(ka, q_frag[kc], st);
+        frag va = *(const frag*)(Vb + (nt * 16 + lo) * DS + kc * 32 + hi * 8);
+        dpt = MF::mma(va, do_frag[kc], dpt);
       }
-      const int kcol = kbase + nt * 16 + lo;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + hi * 4 + r;
+        const int kcol = kbase + nt * 16 + hi * 4 + r;
         float pv = 0.f;
-        if (kcol <= qrow && kcol < S && qrow < S)
-          pv = __expf(st[r] * scale - lse_r[r]);
-        ds[nt][r] = pv * (dpt[r] - dl_r[r]) * scale;
+        if (kcol <= qrow_l && kcol < S && qrow_l < S)
+          pv = __expf(st[r] * scale - lse_q);
+        ds[nt][r] = pv * (dpt[r] - dl_q) * scale;
       }
     }
 
-    // dQ += dS K  (A = dS; B from row-major K via tr_read); KT=64 keys =
-    // two 32-deep contraction chunks.  dS staged ROW-MAJOR [64 key][16 q]
-    // per wave with ONE packed ds_write_b64 per lane per 16-key block
-    // (lane owns 4 consecutive q rows of one key column) instead of 16
-    // scalar ds_write_b16; the A-fragments of dS then come straight out of
-    // the image via ds_read_b64_tr_b16 (PS as in dkdv).
-    constexpr int PS = 20;
-    T* Sw = S_lds + wave * 64 * PS;
+    // dQ += dS K  (A = dS via the packed [16 q][KS] image; B from
+    // row-major K via tr_read); KT=64 keys = two 32-deep contraction chunks
+    T* Sw = S_lds + wave * 16 * KS;
     using P4 = typename Pack4<DT>::V;
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
       P4 s4;
 #pragma unroll
       for (int r = 0; r < 4; ++r) ((T*)&s4)[r] = TR::fromF(ds[nt][r]);
-      *(P4*)(Sw + (nt * 16 + lo) * PS + hi * 4) = s4;
+      *(P4*)(Sw + lo * KS + nt * 16 + hi * 4) = s4;
     }
     frag da[2];
 #pragma unroll
-    for (int c = 0; c < 2; ++c) da[c] = trread_bfrag<DT>(Sw, c * 32, 0, PS, lane);
+    for (int c = 0; c < 2; ++c) da[c] = *(const frag*)(Sw + lo * KS + c * 32 + hi * 8);
 #pragma unroll
     for (int dn = 0; dn < NDN; ++dn)
 #pragma unroll
@@ -1544,8 +1535,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   }
   const int nQT = (int)((S + 127) / 128);    // 8-wave WG: 128 q rows
   const int grid = (int)(B * Hq * nQT);
-  (void)KS2;
-  const size_t lds = sizeof(T) * (4 * KT2 * DS + 8 * 64 * 20);  // dS [64][PS=20]/wave
+  const size_t lds = sizeof(T) * (4 * KT2 * DS + 8 * 16 * KS2);
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,

@@ -1092,4 +1092,42 @@ extern "C" int dk_accum(float* dst, const void* src, int64_t n, int src_dtype,
   return 0;
 }
 
+// ---- split-K dW chunk reduction ----
+// dst[i] += sum_b src[b * chunk_stride + i]: deterministic (fixed summation
+// order over the split-K partial slabs of the weight-gradient GEMM), fp32
+// end to end, fused with the master-grad accumulate — replaces the
+// reference's per-weight cast + autograd-add (grads of F.linear inside
+// transformers Llama, called at train_fsdp.py:383).
+__global__ void accum_chunks_kernel(float* __restrict__ dst,
+                                    const float* __restrict__ src,
+                                    int64_t n, int nchunk, int64_t chunk_stride) {
+  typedef __attribute__((ext_vector_type(4))) float fvec4;
+  const int64_t nv = n / 4;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    fvec4 d = *(const fvec4*)(dst + i * 4);
+    for (int b = 0; b < nchunk; ++b) {
+      fvec4 s = *(const fvec4*)(src + b * chunk_stride + i * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) d[j] += s[j];
+    }
+    *(fvec4*)(dst + i * 4) = d;
+  }
+  for (int64_t i = nv * 4 + blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float d = dst[i];
+    for (int b = 0; b < nchunk; ++b) d += src[b * chunk_stride + i];
+    dst[i] = d;
+  }
+}
+
+extern "C" int dk_accum_chunks(float* dst, const float* src, int64_t n, int nchunk,
+                               int64_t chunk_stride, dkStream stream) {
+  int grid = dk_stream_grid(n / 4 + 1, 256);
+  hipLaunchKernelGGL(accum_chunks_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, dst, src, n, nchunk, chunk_stride);
+  DK_CHECK_LAUNCH();
+  return 0;
+}
+
 extern "C" const char* dk_version(void) { return "diloco_kernels gfx950 0.1.0"; }

@@ -65,13 +65,17 @@ class _CastLinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w16, w_master = ctx.saved_tensors
         dx = dy @ w16
-        dw = (dy.reshape(-1, dy.shape[-1]).t() @ x.reshape(-1, x.shape[-1]))
+        dy2d = dy.reshape(-1, dy.shape[-1])
+        x2d = x.reshape(-1, x.shape[-1])
         wg = w_master.grad
-        if wg is not None and wg.is_cuda and dw.is_cuda:
-            # fused fp32 += bf16 into the flat master grad: one kernel per
-            # weight instead of the cast + autograd-add pair
-            ops._ext().accum_(wg.reshape(-1), dw.reshape(-1))
+        if wg is not None and wg.is_cuda and dy.is_cuda:
+            # split-K dW: batched bf16->fp32 GEMM over token chunks +
+            # deterministic fp32 reduce straight into the flat master grad
+            if ops.dw_splitk_accum(dy2d, x2d, [(wg, 0)]):
+                return dx, None, None
+            ops._ext().accum_(wg.reshape(-1), (dy2d.t() @ x2d).reshape(-1))
             return dx, None, None
+        dw = dy2d.t() @ x2d
         return dx, dw.to(torch.float32), None
 
 
@@ -140,14 +144,23 @@ class _MultiCastLinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, w16cat, *masters = ctx.saved_tensors
         dx = dy @ w16cat
-        dwcat = dy.reshape(-1, dy.shape[-1]).t() @ x.reshape(-1, x.shape[-1])
-        if dwcat.is_cuda and all(m.grad is not None and m.grad.is_cuda for m in masters):
+        dy2d = dy.reshape(-1, dy.shape[-1])
+        x2d = x.reshape(-1, x.shape[-1])
+        if dy.is_cuda and all(m.grad is not None and m.grad.is_cuda for m in masters):
+            targets, off = [], 0
+            for n, m in zip(ctx.sizes, masters):
+                targets.append((m.grad, off))
+                off += n
+            if ops.dw_splitk_accum(dy2d, x2d, targets):
+                return (dx, None, None, *([None] * len(masters)))
             ext = ops._ext()
+            dwcat = dy2d.t() @ x2d
             off = 0
             for n, m in zip(ctx.sizes, masters):
                 ext.accum_(m.grad.reshape(-1), dwcat[off:off + n].reshape(-1))
                 off += n
             return (dx, None, None, *([None] * len(masters)))
+        dwcat = dy2d.t() @ x2d
         dws = []
         off = 0
         for n in ctx.sizes:

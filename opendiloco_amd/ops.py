@@ -425,3 +425,49 @@ class _QKVRopeAttentionFn(torch.autograd.Function):
 
 def qkv_rope_attention(qkv, cos, sin, Hq, Hkv, D, scale):
     return _QKVRopeAttentionFn.apply(qkv, cos, sin, Hq, Hkv, D, scale)
+
+
+# ====================== split-K weight-gradient GEMM ======================
+# dW = dy^T x reduces 32k tokens into a small [N, K] output: a single GEMM
+# yields only ~50-100 workgroups on the 256-CU chip (measured 471-951 TF vs
+# 1300-1475 TF for the fwd/dx GEMM classes — grid starvation).  Split the
+# token dimension into pow2 chunks run as ONE rocBLAS strided-batched
+# bf16->fp32 GEMM, then reduce the fp32 partials deterministically into the
+# fp32 master grads (dk_accum_chunks; numerically stronger than the single
+# bf16-out GEMM — partials never round to bf16).  DK_DW_SPLITK=0 restores
+# the plain-GEMM + accum_ path.
+
+import os as _os
+
+_DW_SPLITK = _os.environ.get("DK_DW_SPLITK", "1") != "0"
+
+
+def _dw_nchunk(R: int, N: int, K: int) -> int:
+    """Pick the split so batch x tiles covers the chip (~768+ workgroups of
+    ~256x128 output tile), capped so chunks keep >= 2048 tokens."""
+    tiles = max(1, (N * K) // (256 * 128))
+    want = 768 // tiles + 1
+    nchunk = 1
+    while nchunk * 2 <= min(want, R // 2048) and R % (nchunk * 2) == 0:
+        nchunk *= 2
+    return nchunk
+
+
+def dw_splitk_accum(dy2d: torch.Tensor, x2d: torch.Tensor,
+                    targets: list[tuple[torch.Tensor, int]]) -> bool:
+    """Accumulate dW = dy2d^T @ x2d into fp32 master grads.
+
+    targets: [(master_grad [n_rows, K] fp32 view, row_offset)] — consecutive
+    row slices of the (possibly concatenated) N dimension.  Returns False if
+    the split-K path is disabled or inapplicable (caller falls back)."""
+    if not _DW_SPLITK or not dy2d.is_cuda:
+        return False
+    R, N = dy2d.shape
+    K = x2d.shape[1]
+    nchunk = _dw_nchunk(R, N, K)
+    ext = _ext()
+    partials = torch.empty(nchunk, N, K, dtype=torch.float32, device=dy2d.device)
+    ext.dw_gemm_batched(dy2d.contiguous(), x2d.contiguous(), partials)
+    for wg, row_off in targets:
+        ext.accum_chunks_(wg.reshape(-1), partials, row_off * K)
+    return True

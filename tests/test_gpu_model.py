@@ -178,3 +178,61 @@ def test_diloco_long_horizon_matches_golden(fixture_2m, golden_dir):
         assert rel < 2e-3, f"step {rec['step']}: {loss_batch} vs {rec['losses'][0]} (rel {rel})"
     assert opt.local_epoch == golden["outer_steps"]
     print(f"worst per-step rel diff over 30 steps: {worst_rel:.2e}")
+
+
+@requires_gpu
+def test_1b_full_depth_matches_golden(golden_dir):
+    """FULL-DEPTH llama-1b (22 layers, GQA 32q/4kv — configs/config_1b.json,
+    BASELINE.json configs[4]) on GPU bf16 vs the fp32 CPU oracle
+    (transformers) golden: 3 DiLoCo steps with one outer crossing, per-step
+    loss within 1e-3 relative, lr exact.  Weights are pinned by construction
+    (oracle/det_init.py: per-parameter crc32-seeded init applied identically
+    to both models — no 4.4 GB fixture)."""
+    from functools import partial
+
+    from oracle.det_init import apply_deterministic_init
+    from opendiloco_amd.data import FakeTokenizedDataLoader
+    from opendiloco_amd.diloco import DiLoCoOptimizer
+    from opendiloco_amd.llama_config import LlamaModelConfig
+    from opendiloco_amd.model import LlamaForCausalLM
+    from opendiloco_amd.optim import clip_grad_norm_flat_
+    from opendiloco_amd.schedule import get_cosine_schedule_with_warmup
+
+    with open(os.path.join(golden_dir, "llama1b_w1_h2_full_depth.json")) as f:
+        golden = json.load(f)
+    cfg = golden["config"]
+
+    mcfg = LlamaModelConfig(vocab_size=32000, hidden_size=2048, intermediate_size=5632,
+                            num_hidden_layers=22, num_attention_heads=32,
+                            num_key_value_heads=4)
+    model = LlamaForCausalLM(mcfg)
+    apply_deterministic_init(model)
+    model = model.to("cuda")
+    model.compute_dtype = torch.bfloat16
+    model.train()
+    opt = DiLoCoOptimizer(
+        batch_size=cfg["batch_size"], num_inner_steps=cfg["local_steps"],
+        outer_optimizer=partial(torch.optim.SGD, lr=cfg["outer_lr"], momentum=0.9, nesterov=True),
+        inner_optimizer=partial(torch.optim.AdamW, lr=cfg["lr"], weight_decay=0.1,
+                                betas=(0.9, 0.95)),
+        params=model.parameters())
+    sched = get_cosine_schedule_with_warmup(opt.inner_optimizer, cfg["warmup_steps"],
+                                            cfg["total_steps"])
+    loader = iter(FakeTokenizedDataLoader(cfg["seq_length"], cfg["vocab_size"],
+                                          cfg["per_device_train_batch_size"], cfg["seed"], 0))
+    grad_acc = cfg["batch_size"] // cfg["per_device_train_batch_size"]
+    for rec in golden["records"]:
+        loss_batch = 0.0
+        for _ in range(grad_acc):
+            batch = {k: v.cuda() for k, v in next(loader).items()}
+            loss = model(**batch).loss / grad_acc
+            loss_batch += loss.item()
+            loss.backward()
+        clip_grad_norm_flat_(opt.flat.flat_grad, 1.0)
+        opt.step()
+        sched.step()
+        opt.zero_grad()
+        assert opt.param_groups[0]["lr"] == rec["lr"], f"step {rec['step']}"
+        rel = abs(loss_batch - rec["losses"][0]) / rec["losses"][0]
+        assert rel < 1e-3, f"step {rec['step']}: bf16 {loss_batch} vs fp32 {rec['losses'][0]} (rel {rel})"
+    assert opt.local_epoch == golden["outer_steps"]
