@@ -53,7 +53,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--model", default="llama-150m", choices=list(MODELS))
     ap.add_argument("--h", type=int, default=50, help="inner steps per outer round")
-    ap.add_argument("--per-device", type=int, default=32)
+    ap.add_argument("--per-device", type=int, default=64)
     ap.add_argument("--batch", type=int, default=512, help="per-worker batch (seqs/step)")
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--precision", default="bf16", choices=["bf16", "fp16"])
