@@ -58,6 +58,9 @@ def main():
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--precision", default="bf16", choices=["bf16", "fp16"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-roofline", action="store_true",
+                    help="skip the in-process roofline/MFMA-fraction measurements "
+                         "(for clean rocprofv3 kernel traces of the step alone)")
     ap.add_argument("--cpu-baseline-only", action="store_true")
     ap.add_argument("--advance-steps", type=int, default=0,
                     help="pre-advance the DiLoCo progress tracker by N inner steps "
@@ -166,7 +169,7 @@ def main():
     # algorithmic traffic = 28 B/param (read p,g,m,v; write p,m,v), HIP events
     # on the launch stream ----
     roofline = None
-    if rank == 0:
+    if rank == 0 and not args.no_roofline:
         from opendiloco_amd.ops import _ext
 
         ext = _ext()

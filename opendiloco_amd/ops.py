@@ -442,9 +442,14 @@ import os as _os
 _DW_SPLITK = _os.environ.get("DK_DW_SPLITK", "1") != "0"
 
 
+_DW_NC_OVERRIDE = _os.environ.get("DK_DW_NC")  # e.g. "8" to force, for sweeps
+
+
 def _dw_nchunk(R: int, N: int, K: int) -> int:
     """Pick the split so batch x tiles covers the chip (~768+ workgroups of
     ~256x128 output tile), capped so chunks keep >= 2048 tokens."""
+    if _DW_NC_OVERRIDE:
+        return min(int(_DW_NC_OVERRIDE), R // 2048)
     tiles = max(1, (N * K) // (256 * 128))
     want = 768 // tiles + 1
     nchunk = 1

@@ -230,3 +230,84 @@ def test_uniform8bit_quantization_roundtrip():
     inside = (buf - offset).abs() <= 127.0 * scale
     assert inside.float().mean() > 0.99
     assert (deq[inside] - buf[inside]).abs().max() <= scale * 0.5 + 1e-7
+
+
+# ---------- 2-process gloo vs oracle golden: BOTH workers pinned ----------
+
+def _golden_worker(rank, world, port, outdir, fixture, golden_path):
+    os.environ.update(dict(RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+                           MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port)))
+    import json
+
+    import torch.distributed as dist
+
+    from opendiloco_amd.data import FakeTokenizedDataLoader
+    from opendiloco_amd.model import LlamaForCausalLM
+    from opendiloco_amd.schedule import get_cosine_schedule_with_warmup
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        with open(golden_path) as f:
+            cfg = json.load(f)["config"]
+        model = LlamaForCausalLM.from_pretrained(fixture).train()
+        opt = DiLoCoOptimizer(
+            batch_size=cfg["batch_size"], num_inner_steps=cfg["local_steps"],
+            outer_optimizer=partial(torch.optim.SGD, lr=cfg["outer_lr"], momentum=0.9,
+                                    nesterov=True),
+            inner_optimizer=partial(torch.optim.AdamW, lr=cfg["lr"], weight_decay=0.1,
+                                    betas=(0.9, 0.95)),
+            params=model.parameters())
+        sched = get_cosine_schedule_with_warmup(opt.inner_optimizer, cfg["warmup_steps"],
+                                                cfg["total_steps"])
+        loader = iter(FakeTokenizedDataLoader(cfg["seq_length"], cfg["vocab_size"],
+                                              cfg["per_device_train_batch_size"],
+                                              cfg["seed"], rank))
+        grad_acc = cfg["batch_size"] // cfg["per_device_train_batch_size"]
+        losses = []
+        for _ in range(cfg["max_steps"]):
+            acc = 0.0
+            for _ in range(grad_acc):
+                batch = next(loader)
+                loss = model(**batch).loss / grad_acc
+                acc += loss.item()
+                loss.backward()
+            torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+            opt.step()
+            sched.step()
+            opt.zero_grad()
+            losses.append(acc)
+        torch.save({"losses": losses, "epoch": opt.local_epoch},
+                   os.path.join(outdir, f"golden_out_{rank}.pt"))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_two_worker_matches_golden_both_ranks(tmp_path, fixture_2m, golden_dir):
+    """Closes the 'worker 0 only' gap: the product stack on 2 gloo ranks must
+    reproduce the oracle golden llama2m_w2_h3 per-step losses for BOTH
+    workers (the CLI/DummyLogger harness only ever sees rank 0's trace)."""
+    import json
+    import socket
+
+    golden_path = os.path.join(golden_dir, "llama2m_w2_h3.json")
+    with socket.socket() as s:
+        s.bind(("", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_golden_worker,
+                         args=(r, 2, port, str(tmp_path), fixture_2m, golden_path))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(600)
+        assert p.exitcode == 0
+    with open(golden_path) as f:
+        golden = json.load(f)
+    outs = [torch.load(tmp_path / f"golden_out_{r}.pt", weights_only=False) for r in range(2)]
+    assert outs[0]["epoch"] == golden["outer_steps"]
+    for i, rec in enumerate(golden["records"]):
+        for r in range(2):
+            got = outs[r]["losses"][i]
+            want = rec["losses"][r]
+            assert abs(got - want) < 1e-3, f"step {rec['step']} worker {r}: {got} vs {want}"
