@@ -444,12 +444,25 @@ _DW_SPLITK = _os.environ.get("DK_DW_SPLITK", "1") != "0"
 
 _DW_NC_OVERRIDE = _os.environ.get("DK_DW_NC")  # e.g. "8" to force, for sweeps
 
+# measured best chunk counts (tools/dw_sweep.py on MI355X, profiles/
+# round2_attn_pmc.md box): (R, N, K) -> nchunk
+_DW_NC_TABLE = {
+    (65536, 3072, 1024): 4, (65536, 1024, 1024): 16, (65536, 5376, 1024): 2,
+    (65536, 1024, 2688): 8, (65536, 32000, 1024): 8,
+    (32768, 3072, 1024): 4, (32768, 1024, 1024): 8, (32768, 5376, 1024): 4,
+    (32768, 1024, 2688): 8, (32768, 32000, 1024): 4,
+}
+
 
 def _dw_nchunk(R: int, N: int, K: int) -> int:
-    """Pick the split so batch x tiles covers the chip (~768+ workgroups of
-    ~256x128 output tile), capped so chunks keep >= 2048 tokens."""
+    """Measured table first; else pick the split so batch x tiles covers the
+    chip (~768+ workgroups of ~256x128 output tile), capped so chunks keep
+    >= 2048 tokens."""
     if _DW_NC_OVERRIDE:
         return min(int(_DW_NC_OVERRIDE), R // 2048)
+    hit = _DW_NC_TABLE.get((R, N, K))
+    if hit is not None:
+        return hit
     tiles = max(1, (N * K) // (256 * 128))
     want = 768 // tiles + 1
     nchunk = 1
