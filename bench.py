@@ -302,18 +302,25 @@ def _mfma_fractions(args, mcfg) -> dict:
               (2 * mcfg.intermediate_size, h, mcfg.num_hidden_layers),
               (h, mcfg.intermediate_size, mcfg.num_hidden_layers),
               (mcfg.vocab_size, h, 1)]
+    from opendiloco_amd.ops import dw_splitk_accum
+
     tot_t = {"fwd": 0.0, "dx": 0.0, "dw": 0.0}
     tot_f = 0.0
     for out_f, in_f, n in shapes:
         x = torch.randn(R, in_f, device="cuda", dtype=dtype)
         w = torch.randn(out_f, in_f, device="cuda", dtype=dtype)
         dy = torch.randn(R, out_f, device="cuda", dtype=dtype)
+        wg = torch.zeros(out_f, in_f, device="cuda", dtype=torch.float32)
         flops = 2.0 * R * out_f * in_f
         tot_f += flops * n
         tot_t["fwd"] += timeit(lambda: F.linear(x, w)) * n
         tot_t["dx"] += timeit(lambda: dy @ w) * n
-        tot_t["dw"] += timeit(lambda: dy.t() @ x) * n
-        del x, w, dy
+        # the dW the model actually runs: split-K batched + fp32 chunk reduce
+        if not dw_splitk_accum(dy, x, [(wg, 0)]):
+            tot_t["dw"] += timeit(lambda: dy.t() @ x) * n
+        else:
+            tot_t["dw"] += timeit(lambda: dw_splitk_accum(dy, x, [(wg, 0)])) * n
+        del x, w, dy, wg
     for kind in tot_t:
         tf = tot_f / tot_t[kind] / 1e12
         out[f"gemm_{kind}"] = {"tflops": tf, "frac": tf / PEAK_TF}
