@@ -66,7 +66,7 @@ def build_binding(force: bool = False) -> str:
         # in the package dir; the absolute PKG_DIR covers the import done from
         # the build dir during this build (and resolves on the GPU box too,
         # where /root/repo is a symlink to the snapshot).
-        extra_ldflags=[f"-L{PKG_DIR}", "-ldiloco_kernels", "-lrocblas", "-Wl,-rpath,$ORIGIN",
+        extra_ldflags=[f"-L{PKG_DIR}", "-ldiloco_kernels", "-lrocblas", "-lhipblaslt", "-Wl,-rpath,$ORIGIN",
                        f"-Wl,-rpath,{PKG_DIR}"],
         build_directory=build_dir,
         is_python_module=False,  # just build; importing is done from the package copy

@@ -11,6 +11,13 @@
 
 #include "../../include/diloco_kernels.h"
 
+#include <rocblas/rocblas.h>
+#include <hipblaslt/hipblaslt.h>
+
+#include <map>
+#include <tuple>
+#include <vector>
+
 namespace {
 
 int dt_of(const at::Tensor& t) {
@@ -302,7 +309,6 @@ std::vector<at::Tensor> attn_bwd_bsd(const at::Tensor& do_bsd, const at::Tensor&
 // fp32 master gradient — numerically STRONGER than the single bf16-out
 // GEMM (partials never round to bf16).  Replaces the reference's
 // F.linear weight-grad + autocast cast (train_fsdp.py:383 grad flow).
-#include <rocblas/rocblas.h>
 
 static rocblas_handle dw_handle() {
   static rocblas_handle h = [] {
@@ -311,6 +317,128 @@ static rocblas_handle dw_handle() {
     return hh;
   }();
   return h;
+}
+
+// hipBLASLt variant of the batched dW GEMM: same math, but the solution is
+// picked by timing the heuristic's top candidates once per shape (a
+// mini-TunableOp; rocBLAS tops out ~1000-1055 TF on these batched
+// bf16->fp32 shapes).  DK_DW_BLASLT=0 falls back to the rocBLAS path.
+
+
+#define LT_OK(call) TORCH_CHECK((call) == HIPBLAS_STATUS_SUCCESS, #call " failed")
+
+static hipblasLtHandle_t lt_handle() {
+  static hipblasLtHandle_t h = [] {
+    hipblasLtHandle_t hh;
+    LT_OK(hipblasLtCreate(&hh));
+    return hh;
+  }();
+  return h;
+}
+
+static at::Tensor& lt_workspace() {
+  static at::Tensor ws;  // lives for the process; torch caching allocator
+  if (!ws.defined())
+    ws = at::empty({64 * 1024 * 1024},
+                   at::TensorOptions().dtype(at::kByte).device(at::kCUDA));
+  return ws;
+}
+
+struct LtPlan {
+  hipblasLtMatmulDesc_t op;
+  hipblasLtMatrixLayout_t la, lb, lc;
+  hipblasLtMatmulAlgo_t algo;
+};
+
+static bool dw_gemm_blaslt(const at::Tensor& dy, const at::Tensor& x,
+                           at::Tensor& partials, int64_t Rc) {
+  static std::map<std::tuple<int64_t, int64_t, int64_t, int64_t>, LtPlan> plans;
+  const int64_t N = dy.size(1), K = x.size(1), nchunk = partials.size(0);
+  const bool is_bf16 = dy.scalar_type() == at::kBFloat16;
+  const hipDataType ab_t = is_bf16 ? HIP_R_16BF : HIP_R_16F;
+  auto key = std::make_tuple(N, K, Rc, nchunk);
+  auto it = plans.find(key);
+  const float alpha = 1.f, beta = 0.f;
+  hipStream_t strm = (hipStream_t)stream();
+  at::Tensor& ws = lt_workspace();
+  const size_t ws_size = (size_t)ws.numel();
+
+  if (it == plans.end()) {
+    LtPlan p;
+    LT_OK(hipblasLtMatmulDescCreate(&p.op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+    hipblasOperation_t ta = HIPBLAS_OP_N, tb = HIPBLAS_OP_T;
+    LT_OK(hipblasLtMatmulDescSetAttribute(p.op, HIPBLASLT_MATMUL_DESC_TRANSA, &ta, sizeof(ta)));
+    LT_OK(hipblasLtMatmulDescSetAttribute(p.op, HIPBLASLT_MATMUL_DESC_TRANSB, &tb, sizeof(tb)));
+    // column-major: C[K, N] = A[K, Rc] (x chunk) * B[N, Rc]^T (dy chunk)
+    auto mk = [&](hipblasLtMatrixLayout_t* l, hipDataType t, int64_t rows,
+                  int64_t cols, int64_t ld, int64_t stride) {
+      LT_OK(hipblasLtMatrixLayoutCreate(l, t, rows, cols, ld));
+      int32_t bc = (int32_t)nchunk;
+      LT_OK(hipblasLtMatrixLayoutSetAttribute(*l, HIPBLASLT_MATRIX_LAYOUT_BATCH_COUNT,
+                                              &bc, sizeof(bc)));
+      int64_t so = stride;
+      LT_OK(hipblasLtMatrixLayoutSetAttribute(
+          *l, HIPBLASLT_MATRIX_LAYOUT_STRIDED_BATCH_OFFSET, &so, sizeof(so)));
+    };
+    mk(&p.la, ab_t, K, Rc, K, Rc * K);
+    mk(&p.lb, ab_t, N, Rc, N, Rc * N);
+    mk(&p.lc, HIP_R_32F, K, N, K, N * K);
+
+    hipblasLtMatmulPreference_t pref;
+    LT_OK(hipblasLtMatmulPreferenceCreate(&pref));
+    uint64_t mws = ws_size;
+    LT_OK(hipblasLtMatmulPreferenceSetAttribute(
+        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &mws, sizeof(mws)));
+    constexpr int kReq = 12;
+    hipblasLtMatmulHeuristicResult_t res[kReq];
+    int nres = 0;
+    hipblasStatus_t hst = hipblasLtMatmulAlgoGetHeuristic(
+        lt_handle(), p.op, p.la, p.lb, p.lc, p.lc, pref, kReq, res, &nres);
+    hipblasLtMatmulPreferenceDestroy(pref);
+    if (hst != HIPBLAS_STATUS_SUCCESS || nres == 0) return false;
+
+    // time each candidate once (3 reps) and keep the fastest
+    float best_ms = 1e30f;
+    int best = -1;
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0);
+    hipEventCreate(&e1);
+    for (int i = 0; i < nres; ++i) {
+      auto run = [&] {
+        return hipblasLtMatmul(lt_handle(), p.op, &alpha, x.data_ptr(), p.la,
+                               dy.data_ptr(), p.lb, &beta, partials.data_ptr(), p.lc,
+                               partials.data_ptr(), p.lc, &res[i].algo,
+                               ws.data_ptr(), ws_size, strm);
+      };
+      if (run() != HIPBLAS_STATUS_SUCCESS) continue;
+      hipEventRecord(e0, strm);
+      for (int r = 0; r < 3; ++r) (void)run();
+      hipEventRecord(e1, strm);
+      hipEventSynchronize(e1);
+      float ms = 0;
+      hipEventElapsedTime(&ms, e0, e1);
+      if (ms < best_ms) { best_ms = ms; best = i; }
+    }
+    hipEventDestroy(e0);
+    hipEventDestroy(e1);
+    if (best < 0) return false;
+    p.algo = res[best].algo;
+    it = plans.emplace(key, p).first;
+  }
+  const LtPlan& p = it->second;
+  hipblasStatus_t st = hipblasLtMatmul(
+      lt_handle(), p.op, &alpha, x.data_ptr(), p.la, dy.data_ptr(), p.lb, &beta,
+      partials.data_ptr(), p.lc, partials.data_ptr(), p.lc, &p.algo,
+      lt_workspace().data_ptr(), ws_size, strm);
+  return st == HIPBLAS_STATUS_SUCCESS;
+}
+
+static bool dw_use_blaslt() {
+  static int v = [] {
+    const char* e = getenv("DK_DW_BLASLT");
+    return (e && e[0] == '0') ? 0 : 1;
+  }();
+  return v != 0;
 }
 
 void dw_gemm_batched(const at::Tensor& dy, const at::Tensor& x, at::Tensor& partials) {
@@ -325,6 +453,7 @@ void dw_gemm_batched(const at::Tensor& dy, const at::Tensor& x, at::Tensor& part
   TORCH_CHECK(x.size(0) == R && partials.size(1) == N && partials.size(2) == K);
   TORCH_CHECK(R % nchunk == 0, "R must divide into nchunk");
   const int64_t Rc = R / nchunk;
+  if (dw_use_blaslt() && dw_gemm_blaslt(dy, x, partials, Rc)) return;
   rocblas_handle h = dw_handle();
   TORCH_CHECK(rocblas_set_stream(h, (hipStream_t)stream()) == rocblas_status_success);
   const rocblas_datatype ab_t = dy.scalar_type() == at::kBFloat16
