@@ -434,9 +434,14 @@ static bool dw_gemm_blaslt(const at::Tensor& dy, const at::Tensor& x,
 }
 
 static bool dw_use_blaslt() {
+  // default OFF: the hipBLASLt algo is picked by runtime timing, so the
+  // solution (and the bitwise result) could vary run to run; the rocBLAS
+  // path with the committed chunk table is deterministic and measured
+  // equal in-step (927 vs 930 ms, tools/dw_sweep A/B).  DK_DW_BLASLT=1
+  // opts in.
   static int v = [] {
     const char* e = getenv("DK_DW_BLASLT");
-    return (e && e[0] == '0') ? 0 : 1;
+    return (e && e[0] == '1') ? 1 : 0;
   }();
   return v != 0;
 }
