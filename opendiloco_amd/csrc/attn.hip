@@ -1199,6 +1199,209 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_v3_kernel(
   }
 }
 
+// ======================= bwd split-32: dv-only / dk-only =======================
+// The fused kernels carry BOTH accumulator sets (dV and dK) and are
+// VGPR-capped at 4 (16x16) / 2 (32x32) waves per SIMD while the matrix
+// pipe idles at ~11% (profiles/round2_attn_pmc.md) — issue/latency bound.
+// Splitting halves the accumulators: each kernel holds ONE floatx16[D/32]
+// set on 32x32x16 tiles (2x the FLOP per instruction of the 16x16 path),
+// recomputes S from lse, and occupies 4-5 waves/SIMD.  QK^T is recomputed
+// by both kernels (+25% matrix-pipe work) against a ~40% cut in issue
+// slots per FLOP.  The mma runs OPERAND-SWAPPED (A/B lane maps coincide)
+// so each lane's C holds 4 consecutive q per key: the P^T / dS^T staging
+// image takes 4 packed ds_write_b64 instead of 16 scalar ds_write_b16 and
+// the lse/delta lookups are float4 reads (same trick as the 16x16 pair).
+// Grid over (b, hq, kv-tile of 128); 4 waves x 32 keys; q-tiles of 32,
+// double-buffered.  Strided dO/V/dV supported (runs in-model, unlike the
+// contiguous-only fused 32x32 port).  WANT_DK=0: dV += P^T dO.
+// WANT_DK=1: dK += dS^T Q with dS^T = P^T o (dP^T - delta) * scale.
+template <int DT, int D, int WANT_DK>
+__global__ __launch_bounds__(256) void attn_bwd_split32_kernel(
+    typename DTraits<DT>::T* __restrict__ out,      // dV or dK
+    const typename DTraits<DT>::T* __restrict__ do_,
+    const typename DTraits<DT>::T* __restrict__ q,
+    const typename DTraits<DT>::T* __restrict__ k,
+    const typename DTraits<DT>::T* __restrict__ v,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    int B, int Hq, int Hkv, int S, float scale,
+    int64_t g_sb, int64_t g_sh, int64_t g_sr,
+    int64_t v_sb, int64_t v_sh, int64_t v_sr,
+    int64_t o_sb, int64_t o_sh, int64_t o_sr) {   // out strides (0 = BHSD)
+  using TR = DTraits<DT>;
+  using T = typename TR::T;
+  using MF = MFMA32<DT>;
+  using frag = typename MF::frag;
+  using P4 = typename Pack4<DT>::V;
+  constexpr int QT = 32;
+  constexpr int QS = QT + 8;
+  constexpr int DS = D + 8;
+  constexpr int NKC = D / 16;
+  constexpr int NMT = D / 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  T* Q_lds = (T*)smem_raw;                   // [2][QT][DS]
+  T* dO_lds = Q_lds + 2 * QT * DS;           // [2][QT][DS]
+  T* PT_lds = dO_lds + 2 * QT * DS;          // [4][32][QS] (P^T or dS^T)
+  float* lse_lds = (float*)(PT_lds + 4 * 32 * QS);   // [2][QT]
+  float* dl_lds = lse_lds + 2 * QT;                  // [2][QT] (dk only)
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo32 = lane & 31;
+  const int hi5 = lane >> 5;
+
+  const int nKT = (S + 127) / 128;
+  int bid = blockIdx.x;
+  const int kt = bid % nKT;
+  const int h = (bid / nKT) % Hq;
+  const int b = bid / (nKT * Hq);
+  const int hkv = h / (Hq / Hkv);
+
+  const int k0 = kt * 128 + wave * 32;
+  const int64_t qoff = (((int64_t)b * Hq + h) * S) * D;
+  const int64_t kvoff = (((int64_t)b * Hkv + hkv) * S) * D;
+  const int64_t lseoff = ((int64_t)b * Hq + h) * S;
+  const int64_t voff = v_sb ? ((int64_t)b * v_sb + (int64_t)hkv * v_sh) : kvoff;
+  const int64_t v_rs = v_sb ? v_sr : (int64_t)D;
+
+  // K (and V for dk) B-fragment base: re-read from L2 per q-tile, freeing
+  // the register copies (v3-port trick)
+  const int kr_c0 = (k0 + lo32) < S ? (k0 + lo32) : S - 1;
+  const T* kbase_p = k + kvoff + (int64_t)kr_c0 * D + hi5 * 8;
+  const T* vbase_p = v + voff + (int64_t)kr_c0 * v_rs + hi5 * 8;
+
+  floatx16 acc[NMT];
+#pragma unroll
+  for (int mt = 0; mt < NMT; ++mt) acc[mt] = (floatx16)(0.f);
+
+  const int qstart = (kt * 128) / QT;
+  const int nQT2 = (S + QT - 1) / QT;
+
+  const int st_t = (int)threadIdx.x;
+  const bool st_on = st_t < (QT * D) / 8;
+  const int st_row = st_t / (D / 8);
+  const int st_c8 = (st_t % (D / 8)) * 8;
+  shortx8 qreg, dreg;
+  float lse_reg = 0.f, dl_reg = 0.f;
+  const int64_t gbase = (int64_t)b * g_sb + (int64_t)h * g_sh;
+
+  auto load_qtile = [&](int qt2) {
+    const int qrow = qt2 * QT + st_row;
+    const int qr_c = qrow < S ? qrow : S - 1;
+    if (st_on) {
+      qreg = *(const shortx8*)(q + qoff + (int64_t)qr_c * D + st_c8);
+      dreg = *(const shortx8*)(do_ + gbase + (int64_t)qr_c * g_sr + st_c8);
+    }
+    if (st_t < QT) {
+      const int rr = qt2 * QT + st_t;
+      const int rr_c = rr < S ? rr : S - 1;
+      lse_reg = lse[lseoff + rr_c];
+      if (WANT_DK) dl_reg = delta[lseoff + rr_c];
+    }
+  };
+  auto write_qtile = [&](int buf) {
+    if (st_on) {
+      *(shortx8*)(Q_lds + buf * QT * DS + st_row * DS + st_c8) = qreg;
+      *(shortx8*)(dO_lds + buf * QT * DS + st_row * DS + st_c8) = dreg;
+    }
+    if (st_t < QT) {
+      lse_lds[buf * QT + st_t] = lse_reg;
+      if (WANT_DK) dl_lds[buf * QT + st_t] = dl_reg;
+    }
+  };
+
+  load_qtile(qstart);
+  write_qtile(0);
+  __syncthreads();
+
+  for (int qt2 = qstart; qt2 < nQT2; ++qt2) {
+    const int qbase = qt2 * QT;
+    const int cur = (qt2 - qstart) & 1;
+    T* Qb = Q_lds + cur * QT * DS;
+    T* dOb = dO_lds + cur * QT * DS;
+    const float* lse_b = lse_lds + cur * QT;
+    const float* dl_b = dl_lds + cur * QT;
+    if (qt2 + 1 < nQT2) load_qtile(qt2 + 1);
+
+    // ---- S (and dP for dk) TRANSPOSED-C: C col = key = lo32,
+    //      row = q = (r&3) + 8*(r>>2) + 4*hi5 ----
+    floatx16 st = (floatx16)(0.f), dpt = (floatx16)(0.f);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kc = 0; kc < NKC; ++kc) {
+      frag qa = *(const frag*)(Qb + lo32 * DS + kc * 16 + hi5 * 8);
+      frag kb = *(const frag*)(kbase_p + kc * 16);
+      st = MF::mma(qa, kb, st);
+      if (WANT_DK) {
+        frag doa = *(const frag*)(dOb + lo32 * DS + kc * 16 + hi5 * 8);
+        frag vb = *(const frag*)(vbase_p + kc * 16);
+        dpt = MF::mma(doa, vb, dpt);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    const int krow = k0 + lo32;
+    const bool full = (qbase >= k0 + 32) && (qbase + QT <= S);
+    // pack P^T (dv) or dS^T (dk) into the [32 key][QS] row-major image,
+    // one ds_write_b64 per 4 consecutive q
+    T* Pw = PT_lds + wave * 32 * QS;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int qloc = 8 * g + 4 * hi5;
+      const floatx4 lse4 = *(const floatx4*)(lse_b + qloc);
+      floatx4 dl4;
+      if (WANT_DK) dl4 = *(const floatx4*)(dl_b + qloc);
+      P4 pk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int i = g * 4 + r;
+        float pv;
+        if (full) {
+          pv = __expf(st[i] * scale - lse4[r]);
+        } else {
+          const int qrow = qbase + qloc + r;
+          pv = 0.f;
+          if (krow <= qrow && krow < S && qrow < S)
+            pv = __expf(st[i] * scale - lse4[r]);
+        }
+        if (WANT_DK) pv = pv * (dpt[i] - dl4[r]) * scale;
+        ((T*)&pk)[r] = TR::fromF(pv);
+      }
+      *(P4*)(Pw + lo32 * QS + qloc) = pk;
+    }
+
+    // ---- acc += A(P^T|dS^T) x B(dO_T|Q_T) ----
+    frag pa[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c)
+      pa[c] = *(const frag*)(Pw + lo32 * QS + c * 16 + hi5 * 8);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        frag bb = trread_afrag32<DT>(WANT_DK ? Qb : dOb, c * 16, mt * 32, DS, lane);
+        acc[mt] = MF::mma(pa[c], bb, acc[mt]);
+      }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (qt2 + 1 < nQT2) write_qtile(cur ^ 1);
+    __syncthreads();
+  }
+
+  // ---- write dV / dK (per q-head; caller sums GQA groups) ----
+  const int64_t obase = o_sb ? ((int64_t)b * o_sb + (int64_t)h * o_sh) : qoff;
+  const int64_t o_rs = o_sb ? o_sr : (int64_t)D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kr = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi5;
+    if (kr >= S) continue;
+#pragma unroll
+    for (int mt = 0; mt < NMT; ++mt)
+      out[obase + (int64_t)kr * o_rs + mt * 32 + lo32] = TR::fromF(acc[mt][r]);
+  }
+}
+
 template <int DT, int D>
 __global__ __launch_bounds__(256) void attn_bwd_dq_v3_kernel(
     typename DTraits<DT>::T* __restrict__ dq_out,
@@ -1377,6 +1580,16 @@ static bool use_attn_v2() {
 // backward defaults to the 16x16 (v2) kernels: measured same-box A/B has
 // them ~7% faster than the 32x32 port (139 vs 148 TF); DK_ATTN_BWD_V3=1
 // opts into the v3 backward for future tuning.
+static bool use_bwd_split32() {
+  // split dv-only/dk-only 32x32 backward (A/B vs the fused 16x16 pair)
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("DK_ATTN_BWD_SPLIT32");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v != 0;
+}
+
 static bool use_bwd_v3() {
   static int cached = -1;
   if (cached < 0) {
@@ -1467,6 +1680,24 @@ static int launch_attn_bwd_dkdv(void* dk_o, void* dv_o, const void* do_, const v
                                 dkStream stream) {
   using T = typename DTraits<DT>::T;
   constexpr int QT = 32, QS = QT + 8, DS = D + 8;
+  if (use_bwd_split32()) {
+    const int nKTs = (int)((S + 127) / 128);
+    const int grids = (int)(B * Hq * nKTs);
+    const size_t ldss = sizeof(T) * (4 * QT * DS + 4 * 32 * QS) + sizeof(float) * 4 * QT;
+    hipLaunchKernelGGL((attn_bwd_split32_kernel<DT, D, 0>), dim3(grids), dim3(256), ldss,
+                       (hipStream_t)stream, (T*)dv_o, (const T*)do_, (const T*)q,
+                       (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
+                       (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr,
+                       dv_sb, dv_sh, dv_sr);
+    DK_CHECK_LAUNCH();
+    hipLaunchKernelGGL((attn_bwd_split32_kernel<DT, D, 1>), dim3(grids), dim3(256), ldss,
+                       (hipStream_t)stream, (T*)dk_o, (const T*)do_, (const T*)q,
+                       (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
+                       (int)Hkv, (int)S, scale, g_sb, g_sh, g_sr, v_sb, v_sh, v_sr,
+                       0, 0, 0);
+    DK_CHECK_LAUNCH();
+    return 0;
+  }
   if (use_bwd_v3() && v_sb == 0 && dv_sb == 0) {  // v3 port is contiguous-only
     const int nKT3 = (int)((S + 127) / 128);
     const int grid3 = (int)(B * Hq * nKT3);
