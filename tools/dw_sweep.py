@@ -13,8 +13,10 @@ sys.path.insert(0, REPO_ROOT)
 
 import torch  # noqa: E402
 
-SHAPES = [("qkv", 3072, 1024), ("o", 1024, 1024), ("gateup", 5376, 1024),
-          ("down", 1024, 2688), ("lm_head", 32000, 1024)]
+SHAPES_150M = [("qkv", 3072, 1024), ("o", 1024, 1024), ("gateup", 5376, 1024),
+               ("down", 1024, 2688), ("lm_head", 32000, 1024)]
+SHAPES_1B = [("qkv", 2560, 2048), ("o", 2048, 2048), ("gateup", 11264, 2048),
+             ("down", 2048, 5632), ("lm_head", 32000, 2048)]
 
 
 def _timeit(fn, reps=10):
@@ -33,7 +35,9 @@ def _timeit(fn, reps=10):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--r", type=int, default=65536)
+    ap.add_argument("--model", default="150m", choices=["150m", "1b"])
     args = ap.parse_args()
+    SHAPES = SHAPES_150M if args.model == "150m" else SHAPES_1B
     from opendiloco_amd import ops
 
     ext = ops._ext()
