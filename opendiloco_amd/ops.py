@@ -451,6 +451,11 @@ _DW_NC_TABLE = {
     (65536, 1024, 2688): 8, (65536, 32000, 1024): 8,
     (32768, 3072, 1024): 4, (32768, 1024, 1024): 8, (32768, 5376, 1024): 4,
     (32768, 1024, 2688): 8, (32768, 32000, 1024): 4,
+    # llama-1b shapes (GQA qkv 2560 = 2048 q + 2x256 kv)
+    (16384, 2560, 2048): 2, (16384, 2048, 2048): 4, (16384, 11264, 2048): 2,
+    (16384, 2048, 5632): 1, (16384, 32000, 2048): 2,
+    (32768, 2560, 2048): 4, (32768, 2048, 2048): 4, (32768, 11264, 2048): 8,
+    (32768, 2048, 5632): 4, (32768, 32000, 2048): 4,
 }
 
 
