@@ -4,6 +4,8 @@ PyTorch fp32 reference of the same op, on the same (bf16-rounded) inputs.
 Run: gpurun -- 'python -m pytest tests -m gpu -x -q'
 """
 
+import os
+
 import pytest
 import torch
 
@@ -473,3 +475,50 @@ def test_dw_splitk_accum_matches_fp32(ext, N, K, slices):
     assert ops.dw_splitk_accum(dy, x, targets2)
     for a, b in zip(masters, m2):
         assert torch.equal(a, b)
+
+
+@requires_gpu
+@pytest.mark.parametrize("env", [{"DK_ATTN_V2": "1"}, {"DK_ATTN_BWD_V3": "1"},
+                                 {"DK_ATTN_BWD_SPLIT32": "1"}])
+def test_attention_env_variants_parity(repo_root, env):
+    """The env-gated kernel variants (16x16 forward, fused 32x32 backward,
+    split dv/dk 32x32 backward — kept as measured A/B alternatives,
+    DESIGN.md §3) must stay numerically correct.  Env is latched at first
+    use, so each variant runs in a subprocess."""
+    import subprocess
+    import sys
+
+    code = """
+import torch
+from opendiloco_amd.ops import _ext
+ext = _ext()
+torch.manual_seed(3)
+B, Hq, S, D = 4, 4, 512, 64
+q = torch.randn(B, Hq, S, D, device="cuda", dtype=torch.bfloat16)
+k, v = torch.randn_like(q), torch.randn_like(q)
+scale = D ** -0.5
+o, lse = ext.attn_fwd(q, k, v, scale)
+do = torch.randn_like(o)
+dq, dk, dv = ext.attn_bwd(do, q, k, v, o, lse, scale)
+# fp32 torch reference
+qf, kf, vf = q.float(), k.float(), v.float()
+for t in (qf, kf, vf):
+    t.requires_grad_(True)
+mask = torch.ones(S, S, device="cuda", dtype=torch.bool).tril()
+s = (qf @ kf.transpose(-1, -2)) * scale
+s = s.masked_fill(~mask, float("-inf"))
+of = torch.softmax(s, dim=-1) @ vf
+assert (o.float() - of).abs().max() < 2e-2, "fwd"
+of.backward(do.float())
+for got, want, name in ((dq, qf.grad, "dq"), (dk, kf.grad, "dk"), (dv, vf.grad, "dv")):
+    err = (got.float() - want).abs().max().item()
+    assert err < 5e-2, f"{name} err {err}"
+print("variant ok")
+"""
+    e = dict(os.environ)
+    e.update(env)
+    e["PYTHONPATH"] = repo_root + os.pathsep + e.get("PYTHONPATH", "")
+    r = subprocess.run([sys.executable, "-c", code], env=e, capture_output=True,
+                       text=True, timeout=600, cwd=repo_root)
+    assert r.returncode == 0, f"{env}: {r.stderr[-2000:]}"
+    assert "variant ok" in r.stdout
