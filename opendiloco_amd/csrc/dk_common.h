@@ -19,11 +19,15 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 }
 
 __device__ __forceinline__ unsigned short f32_to_bf16(float f) {
-  union { float f; unsigned int i; } c;
-  c.f = f;
-  if ((c.i & 0x7fffffffu) > 0x7f800000u) return 0x7fc0u;  // NaN
-  unsigned int lsb = (c.i >> 16) & 1u;
-  return (unsigned short)((c.i + 0x7fffu + lsb) >> 16);   // round-to-nearest-even
+  // native cast lowers to ONE v_cvt_pk_bf16_f32 (RTNE on gfx950; adjacent
+  // casts pair into a single pk instruction).  The previous manual
+  // round-to-nearest-even (+0x7fff+lsb shift chain, ~6 VALU) cost the
+  // issue-bound attention kernels 15-25% — measured as the f16-vs-bf16
+  // gap in profiles/round2: identical kernels, v_cvt_f16 vs manual bf16.
+  __bf16 h = (__bf16)f;
+  unsigned short u;
+  __builtin_memcpy(&u, &h, 2);
+  return u;
 }
 
 // ---------- dtype traits: DT 0=f32, 1=f16, 2=bf16 ----------
