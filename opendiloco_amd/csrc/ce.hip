@@ -46,11 +46,9 @@ __global__ void ce_fwd_kernel(float* __restrict__ loss_rows, float* __restrict__
       // per-vector max first, then one rescale + W independent exps: breaks
       // the per-element serial rescale chain (latency-bound) into ILP
       float xf[W], vm = -INFINITY;
+      packed_to_f32<DT, W>(&xv, xf);
 #pragma unroll
-      for (int j = 0; j < W; ++j) {
-        xf[j] = TR::toF(((const TT*)&xv)[j]);
-        vm = fmaxf(vm, xf[j]);
-      }
+      for (int j = 0; j < W; ++j) vm = fmaxf(vm, xf[j]);
       if (vm > m) { s *= __expf(m - vm); m = vm; }
       float ps = 0.f;
 #pragma unroll
@@ -116,10 +114,12 @@ __global__ void ce_bwd_kernel(typename DTraits<DT>::T* __restrict__ dlogits,
     for (int64_t i = threadIdx.x; i < nvec; i += NT) {
       V8 xv = *(const V8*)(row + i * W);
       V8 dv;
+      float xf[W];
+      packed_to_f32<DT, W>(&xv, xf);
 #pragma unroll
       for (int j = 0; j < W; ++j) {
         int64_t v = i * W + j;
-        float p = __expf(TR::toF(((const TT*)&xv)[j]) - l);
+        float p = __expf(xf[j] - l);
         float d = (p - (v == lab ? 1.f : 0.f)) * scale;
         ((TT*)&dv)[j] = TR::fromF(d);
       }

@@ -120,8 +120,7 @@ __global__ void rmsnorm_fwd_vec_kernel(typename DTraits<DT>::T* __restrict__ y,
   float wf[VEC];
   {
     vec_t wvv = *(const vec_t*)(w + c0);
-#pragma unroll
-    for (int j = 0; j < VEC; ++j) wf[j] = TR::toF(((const T*)&wvv)[j]);
+    packed_to_f32<DT, VEC>(&wvv, wf);
   }
   // software pipeline: next row's loads are issued before this row's
   // block-reduce barrier so HBM latency overlaps the sync
@@ -134,20 +133,21 @@ __global__ void rmsnorm_fwd_vec_kernel(typename DTraits<DT>::T* __restrict__ y,
     float hf[VEC], ss = 0.f;
     if (RES) {
       vec_t hv;
+      float xf[VEC], rf[VEC];
+      packed_to_f32<DT, VEC>(&xv, xf);
+      packed_to_f32<DT, VEC>(&rv, rf);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
-        T h = TR::fromF(TR::toF(((const T*)&xv)[j]) + TR::toF(((const T*)&rv)[j]));
+        T h = TR::fromF(xf[j] + rf[j]);
         ((T*)&hv)[j] = h;
         hf[j] = TR::toF(h);
         ss += hf[j] * hf[j];
       }
       *(vec_t*)(h_out + r * cols + c0) = hv;
     } else {
+      packed_to_f32<DT, VEC>(&xv, hf);
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) {
-        hf[j] = TR::toF(((const T*)&xv)[j]);
-        ss += hf[j] * hf[j];
-      }
+      for (int j = 0; j < VEC; ++j) ss += hf[j] * hf[j];
     }
     if (r + gridDim.x < rows) {
       xv = *(const vec_t*)(x + (r + gridDim.x) * cols + c0);
@@ -269,8 +269,9 @@ __global__ void rmsnorm_bwd_vec_kernel(typename DTraits<DT>::T* __restrict__ dx,
   float wv[VEC], dwacc[VEC];
   {
     vec_t wvv = *(const vec_t*)(w + c0);
+    packed_to_f32<DT, VEC>(&wvv, wv);
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) { wv[j] = TR::toF(((const T*)&wvv)[j]); dwacc[j] = 0.f; }
+    for (int j = 0; j < VEC; ++j) dwacc[j] = 0.f;
   }
   vec_t xv, dv, rv;
   if (blockIdx.x < rows) {
@@ -281,12 +282,9 @@ __global__ void rmsnorm_bwd_vec_kernel(typename DTraits<DT>::T* __restrict__ dx,
   for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
     const float ir = invrms[r];
     float xf[VEC], dyf[VEC], drf[VEC];
-#pragma unroll
-    for (int j = 0; j < VEC; ++j) {
-      xf[j] = TR::toF(((const T*)&xv)[j]);
-      dyf[j] = TR::toF(((const T*)&dv)[j]);
-      if (DRES) drf[j] = TR::toF(((const T*)&rv)[j]);
-    }
+    packed_to_f32<DT, VEC>(&xv, xf);
+    packed_to_f32<DT, VEC>(&dv, dyf);
+    if (DRES) packed_to_f32<DT, VEC>(&rv, drf);
     if (r + gridDim.x < rows) {
       xv = *(const vec_t*)(x + (r + gridDim.x) * cols + c0);
       dv = *(const vec_t*)(dy + (r + gridDim.x) * cols + c0);
