@@ -514,10 +514,13 @@ __global__ void rope_move_kernel(typename DTraits<DT>::T* __restrict__ out,
       const float4 sv0 = *(const float4*)(sintab + (int64_t)sp * hd + i0);
       const float4 sv1 = *(const float4*)(sintab + (int64_t)sp * hd + i0 + 4);
       shortx8 o1, o2;
+      float af[8], bf[8];
+      packed_to_f32<DT, 8>(&x1, af);
+      packed_to_f32<DT, 8>(&x2, bf);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float a = TR::toF(((const T*)&x1)[j]);
-        float bb = TR::toF(((const T*)&x2)[j]);
+        float a = af[j];
+        float bb = bf[j];
         float c = j < 4 ? ((const float*)&cv0)[j] : ((const float*)&cv1)[j - 4];
         float sn = j < 4 ? ((const float*)&sv0)[j] : ((const float*)&sv1)[j - 4];
         if (BWD) sn = -sn;

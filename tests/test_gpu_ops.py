@@ -158,6 +158,7 @@ def test_cross_entropy_fwd_bwd(ext, B, S, V):
     (1, 16, 16, 1024, 64),  # 150m full seq
     (2, 8, 2, 128, 64),     # GQA (1b shape)
     (1, 2, 2, 96, 64),      # ragged tail (S % 64 != 0)
+    (1, 2, 2, 640, 64),     # S % 128 != 0 with multiple kv tiles (8-wave WG tail)
 ])
 def test_attention_fwd_bwd(ext, B, Hq, Hkv, S, D):
     from opendiloco_amd import ops
