@@ -316,10 +316,17 @@ def _mfma_fractions(args, mcfg) -> dict:
         tot_t["fwd"] += timeit(lambda: F.linear(x, w)) * n
         tot_t["dx"] += timeit(lambda: dy @ w) * n
         # the dW the model actually runs: split-K batched + fp32 chunk reduce
+        # (dw_stream_sync so the side-stream work is inside the timed region)
+        from opendiloco_amd.ops import dw_stream_sync
+
         if not dw_splitk_accum(dy, x, [(wg, 0)]):
             tot_t["dw"] += timeit(lambda: dy.t() @ x) * n
         else:
-            tot_t["dw"] += timeit(lambda: dw_splitk_accum(dy, x, [(wg, 0)])) * n
+            def _dw():
+                dw_splitk_accum(dy, x, [(wg, 0)])
+                dw_stream_sync()
+
+            tot_t["dw"] += timeit(_dw) * n
         del x, w, dy, wg
     for kind in tot_t:
         tf = tot_f / tot_t[kind] / 1e12

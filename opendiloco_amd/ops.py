@@ -476,6 +476,39 @@ def _dw_nchunk(R: int, N: int, K: int) -> int:
     return nchunk
 
 
+# ---- side-stream dW (overlap with the rest of backward) ----
+# The dW GEMMs are compute-dense and chip-filling while the attention
+# backward (which dominates the rest of the backward wall) keeps the
+# matrix pipes ~94% idle (profiles/round2_attn_pmc.md): running dW on a
+# side HIP stream lets layer L's weight gradients overlap layers L-1..1's
+# backward.  Determinism is preserved: each master-grad buffer is only
+# ever written from the side stream, in a fixed order.  Stream ordering
+# for grad consumers is restored by (a) an end-of-backward autograd
+# callback that makes the main stream wait on the side stream, and (b) an
+# explicit dw_stream_sync() in clip_grad_norm_flat_.  DK_DW_ASYNC=0
+# disables the overlap (dW runs inline on the current stream).
+_DW_ASYNC = _os.environ.get("DK_DW_ASYNC", "1") != "0"
+_dw_stream_box: list = []
+_dw_sync_task = [-2]  # graph-task id the sync callback is queued for
+
+
+def _dw_stream():
+    if not _dw_stream_box:
+        _dw_stream_box.append(torch.cuda.Stream())
+    return _dw_stream_box[0]
+
+
+def dw_stream_sync() -> None:
+    """Make the current stream wait for all queued side-stream dW work."""
+    if _dw_stream_box:
+        torch.cuda.current_stream().wait_stream(_dw_stream_box[0])
+
+
+def _dw_sync_cb():
+    _dw_sync_task[0] = -2
+    dw_stream_sync()
+
+
 def dw_splitk_accum(dy2d: torch.Tensor, x2d: torch.Tensor,
                     targets: list[tuple[torch.Tensor, int]]) -> bool:
     """Accumulate dW = dy2d^T @ x2d into fp32 master grads.
@@ -489,8 +522,40 @@ def dw_splitk_accum(dy2d: torch.Tensor, x2d: torch.Tensor,
     K = x2d.shape[1]
     nchunk = _dw_nchunk(R, N, K)
     ext = _ext()
+    dy2d = dy2d.contiguous()
+    x2d = x2d.contiguous()
+    if _DW_ASYNC:
+        side = _dw_stream()
+        cur = torch.cuda.current_stream()
+        ev = torch.cuda.Event()
+        ev.record(cur)
+        with torch.cuda.stream(side):
+            side.wait_event(ev)
+            partials = torch.empty(nchunk, N, K, dtype=torch.float32, device=dy2d.device)
+            ext.dw_gemm_batched(dy2d, x2d, partials)
+            for wg, row_off in targets:
+                ext.accum_chunks_(wg.reshape(-1), partials, row_off * K)
+        # keep the inputs alive until the side stream is done with them
+        dy2d.record_stream(side)
+        x2d.record_stream(side)
+        # restore stream ordering for anyone reading .grad after backward():
+        # queue ONE end-of-backward callback per autograd graph task (the id
+        # also distinguishes a pass aborted by an exception from a new one)
+        try:
+            tid = torch._C._current_graph_task_id()
+        except AttributeError:
+            tid = -1
+        if tid == -1:
+            dw_stream_sync()  # not inside a backward pass (e.g. bench timing)
+        elif _dw_sync_task[0] != tid:
+            try:
+                torch.autograd.Variable._execution_engine.queue_callback(_dw_sync_cb)
+                _dw_sync_task[0] = tid
+            except Exception:
+                dw_stream_sync()
+        return True
     partials = torch.empty(nchunk, N, K, dtype=torch.float32, device=dy2d.device)
-    ext.dw_gemm_batched(dy2d.contiguous(), x2d.contiguous(), partials)
+    ext.dw_gemm_batched(dy2d, x2d, partials)
     for wg, row_off in targets:
         ext.accum_chunks_(wg.reshape(-1), partials, row_off * K)
     return True

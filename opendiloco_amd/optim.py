@@ -233,8 +233,9 @@ def clip_grad_norm_flat_(flat_grad: torch.Tensor, max_norm: float) -> torch.Tens
     at train_fsdp.py:395 / train_diloco_torch.py:323).  GPU: deterministic
     two-pass HIP kernels; CPU: torch."""
     if flat_grad.device.type == "cuda":
-        from opendiloco_amd.ops import _ext
+        from opendiloco_amd.ops import _ext, dw_stream_sync
 
+        dw_stream_sync()  # side-stream dW accumulation must land first
         out2 = _ext().clip_grad_(flat_grad, max_norm)
         return out2[0]
     total = flat_grad.norm(2)

@@ -523,3 +523,47 @@ print("variant ok")
                        text=True, timeout=600, cwd=repo_root)
     assert r.returncode == 0, f"{env}: {r.stderr[-2000:]}"
     assert "variant ok" in r.stdout
+
+
+@requires_gpu
+def test_dw_async_stream_matches_sync(repo_root):
+    """Side-stream dW overlap (DK_DW_ASYNC, default on) must be BIT-equal
+    to the inline path: same accumulation order per buffer, and the
+    end-of-backward callback restores stream ordering so grads read right
+    after backward() are complete."""
+    import subprocess
+    import sys
+
+    code = """
+import torch
+from opendiloco_amd.llama_config import LlamaModelConfig
+from opendiloco_amd.model import LlamaForCausalLM
+torch.manual_seed(0)
+cfg = LlamaModelConfig(vocab_size=1024, hidden_size=128, intermediate_size=256,
+                       num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=2)
+m = LlamaForCausalLM(cfg).init_weights(seed=11).to("cuda")
+m.compute_dtype = torch.bfloat16
+ids = torch.randint(3, 1024, (4, 256), device="cuda")
+b = dict(input_ids=ids, attention_mask=torch.ones_like(ids), labels=ids.clone())
+for p in m.parameters():
+    p.grad = torch.zeros_like(p)
+m(**b).loss.backward()
+# read grads IMMEDIATELY (the end-of-backward callback must have ordered them)
+h = 0.0
+for n, p in sorted(m.named_parameters()):
+    h += p.grad.double().abs().sum().item()
+import hashlib
+sig = hashlib.md5(b"".join(p.grad.float().cpu().numpy().tobytes()
+                           for _, p in sorted(m.named_parameters()))).hexdigest()
+print("SIG", sig, h)
+"""
+    outs = []
+    for async_flag in ("1", "0"):
+        e = dict(os.environ)
+        e["DK_DW_ASYNC"] = async_flag
+        e["PYTHONPATH"] = repo_root + os.pathsep + e.get("PYTHONPATH", "")
+        r = subprocess.run([sys.executable, "-c", code], env=e, capture_output=True,
+                           text=True, timeout=600, cwd=repo_root)
+        assert r.returncode == 0, f"async={async_flag}: {r.stderr[-2000:]}"
+        outs.append([l for l in r.stdout.splitlines() if l.startswith("SIG")][0])
+    assert outs[0] == outs[1], f"async vs sync grads differ: {outs}"
