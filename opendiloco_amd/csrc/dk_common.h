@@ -124,6 +124,25 @@ __device__ __forceinline__ float block_reduce_sum(float x, float* smem) {
   return r;
 }
 
+// two independent sums through ONE barrier (amortizes the per-row block
+// sync of row-per-block kernels when two rows are processed per iteration);
+// smem: float[2 * NT/64]
+template <int NT>
+__device__ __forceinline__ void block_reduce_sum2(float& a, float& b, float* smem) {
+  constexpr int NW = NT / DK_WAVE;
+  const int wid = threadIdx.x / DK_WAVE;
+  a = wave_reduce_sum(a);
+  b = wave_reduce_sum(b);
+  if ((threadIdx.x & (DK_WAVE - 1)) == 0) { smem[wid] = a; smem[NW + wid] = b; }
+  __syncthreads();
+  float ra = 0.f, rb = 0.f;
+#pragma unroll
+  for (int i = 0; i < NW; ++i) { ra += smem[i]; rb += smem[NW + i]; }
+  __syncthreads();
+  a = ra;
+  b = rb;
+}
+
 template <int NT>
 __device__ __forceinline__ float block_reduce_max(float x, float* smem) {
   constexpr int NW = NT / DK_WAVE;
