@@ -849,9 +849,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   constexpr int NDN = D / 16;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  T* K_lds = (T*)smem_raw;                    // [2][KT][DS]
-  T* V_lds = K_lds + 2 * KT * DS;             // [2][KT][DS]
-  T* S_lds = V_lds + 2 * KT * DS;             // [8][16][KS]  (dQ B-frags come
+  T* K_lds = (T*)smem_raw;                    // [3][KT][DS] (triple ring)
+  T* V_lds = K_lds + 3 * KT * DS;             // [3][KT][DS]
+  T* S_lds = V_lds + 3 * KT * DS;             // [8][16][KS]  (dQ B-frags come
                                               //  from row-major K via tr_read)
 
   const int lane = threadIdx.x & 63;
@@ -895,38 +895,47 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   const int kv_end = min(S, qt * 128 + 128);
   const int n_kt = (kv_end + KT - 1) / KT;
 
-  // async double-buffered staging (one piece per thread: KT*D/8 <= 256)
+  // async TRIPLE-buffered staging (prefetch distance 2: the t+2 load is
+  // issued at the START of tile t, the t+1 registers land in LDS at the
+  // END of tile t — ~2 compute tiles of load slack; the double-buffered
+  // scheme gave <1 and left dq 50% parked, round2_attn_pmc.md).  Two
+  // register sets (A = even tiles, B = odd) cost +8 VGPRs (114 total,
+  // still 4 waves/SIMD); 3 LDS buffers fit 2 WGs/CU (147 KB).
   const int st_t = (int)threadIdx.x;
   const bool st_on = st_t < (KT * D) / 8;
   const int st_row = st_t / (D / 8);
   const int st_c8 = (st_t % (D / 8)) * 8;
-  frag kreg, vreg;
+  frag kregA, vregA, kregB, vregB;
 
-  auto load_ktile = [&](int kt) {
+  auto load_ktile = [&](int kt, frag& kr, frag& vr) {
     if (st_on) {
       const int krow = kt * KT + st_row;
       const int kr_c = krow < S ? krow : S - 1;
-      kreg = *(const frag*)(k + kvoff + (int64_t)kr_c * D + st_c8);
-      vreg = *(const frag*)(v + voff + (int64_t)kr_c * v_rs + st_c8);
+      kr = *(const frag*)(k + kvoff + (int64_t)kr_c * D + st_c8);
+      vr = *(const frag*)(v + voff + (int64_t)kr_c * v_rs + st_c8);
     }
   };
-  auto write_ktile = [&](int buf) {
+  auto write_ktile = [&](int buf, const frag& kr, const frag& vr) {
     if (st_on) {
-      *(frag*)(K_lds + buf * KT * DS + st_row * DS + st_c8) = kreg;
-      *(frag*)(V_lds + buf * KT * DS + st_row * DS + st_c8) = vreg;
+      *(frag*)(K_lds + buf * KT * DS + st_row * DS + st_c8) = kr;
+      *(frag*)(V_lds + buf * KT * DS + st_row * DS + st_c8) = vr;
     }
   };
 
-  load_ktile(0);
-  write_ktile(0);
+  load_ktile(0, kregA, vregA);
+  write_ktile(0, kregA, vregA);
+  if (1 < n_kt) load_ktile(1, kregB, vregB);
   __syncthreads();
 
   for (int kt = 0; kt < n_kt; ++kt) {
     const int kbase = kt * KT;
-    const int cur = kt & 1;
+    const int cur = kt % 3;
     T* Kb = K_lds + cur * KT * DS;
     T* Vb = V_lds + cur * KT * DS;
-    if (kt + 1 < n_kt) load_ktile(kt + 1);
+    if (kt + 2 < n_kt) {
+      if ((kt + 2) & 1) load_ktile(kt + 2, kregB, vregB);
+      else load_ktile(kt + 2, kregA, vregA);
+    }
 
     // wave-uniform diagonal skip (8-wave skew): if every key in this kv
     // tile exceeds this wave's last q row, the whole tile is masked to zero.
@@ -983,7 +992,10 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     __builtin_amdgcn_s_setprio(0);
     }  // end diagonal skip
 
-    if (kt + 1 < n_kt) write_ktile(cur ^ 1);  // T14: write late
+    if (kt + 1 < n_kt) {  // T14: write late (registers loaded last iteration)
+      if ((kt + 1) & 1) write_ktile((kt + 1) % 3, kregB, vregB);
+      else write_ktile((kt + 1) % 3, kregA, vregA);
+    }
     __syncthreads();
   }
 
@@ -1766,7 +1778,7 @@ static int launch_attn_bwd_dq(void* dq_o, const void* do_, const void* q, const 
   }
   const int nQT = (int)((S + 127) / 128);    // 8-wave WG: 128 q rows
   const int grid = (int)(B * Hq * nQT);
-  const size_t lds = sizeof(T) * (4 * KT2 * DS + 8 * 16 * KS2);
+  const size_t lds = sizeof(T) * (6 * KT2 * DS + 8 * 16 * KS2);  // 3-ring K+V
   hipLaunchKernelGGL((attn_bwd_dq_kernel<DT, D>), dim3(grid), dim3(512), lds,
                      (hipStream_t)stream, (T*)dq_o, (const T*)do_, (const T*)q,
                      (const T*)k, (const T*)v, lse, delta, (int)B, (int)Hq,
