@@ -897,10 +897,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
 
   // async TRIPLE-buffered staging (prefetch distance 2: the t+2 load is
   // issued at the START of tile t, the t+1 registers land in LDS at the
-  // END of tile t — ~2 compute tiles of load slack; the double-buffered
-  // scheme gave <1 and left dq 50% parked, round2_attn_pmc.md).  Two
-  // register sets (A = even tiles, B = odd) cost +8 VGPRs (114 total,
-  // still 4 waves/SIMD); 3 LDS buffers fit 2 WGs/CU (147 KB).
+  // END of tile t — ~2 compute tiles of load slack).  Two register sets
+  // (A = even tiles, B = odd) cost +8 VGPRs (120 total, still 4
+  // waves/SIMD); 3 LDS buffers fit 2 WGs/CU (147 KB).  MEASURED: a wash
+  // vs double-buffering (209 vs 212 TF) — dq's 50% parked
+  // (round2_attn_pmc.md) is LDS-read latency inside the mma loops and
+  // barrier convoy, not global staging slack.  Kept: equal speed,
+  // deeper slack for other shapes.
   const int st_t = (int)threadIdx.x;
   const bool st_on = st_t < (KT * D) / 8;
   const int st_row = st_t / (D / 8);
