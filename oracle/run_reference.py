@@ -68,6 +68,13 @@ TRACE_CASES = {
         local_steps=1, max_steps=6, lr=4e-4, outer_lr=0.7,
         warmup_steps=1000, total_steps=88_000, seed=42, vocab_size=1024,
     ),
+    # the reference's own e2e test shape (test_train.py:24-39: batch 16/8,
+    # seq 1024) at H=5 — matches tests/golden/llama2m_w2_h5_seq1024.json
+    "reference_trace_w2_h5_seq1024": dict(
+        nproc=2, batch_size=16, per_device_train_batch_size=8, seq_length=1024,
+        local_steps=5, max_steps=10, lr=1e-2, outer_lr=0.7,
+        warmup_steps=1000, total_steps=88_000, seed=42, vocab_size=1024,
+    ),
 }
 
 

@@ -40,6 +40,7 @@ def test_oracle_matches_golden(golden_dir, fixture_2m, name):
 @pytest.mark.parametrize("trace_name,n_workers,local_steps", [
     ("reference_trace_w1_h1", 1, 1),
     ("reference_trace_w2_h3", 2, 3),
+    ("reference_trace_w2_h5_seq1024", 2, 5),  # the reference's e2e test shape
 ])
 def test_oracle_matches_reference_executed_trace(golden_dir, fixture_2m, trace_name,
                                                  n_workers, local_steps):
@@ -53,16 +54,19 @@ def test_oracle_matches_reference_executed_trace(golden_dir, fixture_2m, trace_n
     trace = _load_golden(golden_dir, trace_name)
     tc = trace["config"]
     assert tc["nproc"] == n_workers and tc["local_steps"] == local_steps
+    # cap the re-run at one outer crossing so the CPU suite stays fast
+    # (the committed trace still holds the full record count)
+    max_steps = min(tc["max_steps"], local_steps + 1)
     cfg = OracleConfig(model_path=fixture_2m, n_workers=n_workers,
                        local_steps=local_steps, batch_size=tc["batch_size"],
                        per_device_train_batch_size=tc["per_device_train_batch_size"],
-                       seq_length=tc["seq_length"], max_steps=tc["max_steps"],
+                       seq_length=tc["seq_length"], max_steps=max_steps,
                        lr=tc["lr"], outer_lr=tc["outer_lr"],
                        warmup_steps=tc["warmup_steps"], total_steps=tc["total_steps"],
                        seed=tc["seed"], vocab_size=tc["vocab_size"])
     result = run_diloco_oracle(cfg)
-    assert len(result["records"]) == len(trace["records"])
-    for got, want in zip(result["records"], trace["records"]):
+    assert len(result["records"]) == max_steps
+    for got, want in zip(result["records"], trace["records"][:max_steps]):
         assert got["step"] == want["step"]
         assert got["lr"] == want["lr"], f"lr mismatch at step {got['step']}"
         # worker 0's loss <-> the reference's rank-0 wandb "Loss"
