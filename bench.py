@@ -118,13 +118,25 @@ def main():
         micros.append(dict(input_ids=ids, attention_mask=torch.ones_like(ids),
                            labels=ids.clone()))
 
+    # run the main compute on a HIGH-priority stream: the side-stream dW
+    # GEMMs (ops.py DK_DW_ASYNC) otherwise starve the main stream's small
+    # kernels at workgroup-dispatch arbitration (measured: rmsnorm's reduce
+    # stretched 12 -> 184 us under overlap).  DK_MAIN_PRIO=0 disables.
+    main_stream = (torch.cuda.Stream(priority=-1)
+                   if os.environ.get("DK_MAIN_PRIO", "1") != "0" else None)
+
     def one_step():
-        for mb in micros:
-            loss = model(**mb).loss / grad_acc
-            loss.backward()
-        clip_grad_norm_flat_(opt.flat.flat_grad, 1.0)
-        opt.step()
-        opt.zero_grad()
+        import contextlib
+
+        with torch.cuda.stream(main_stream) if main_stream else contextlib.nullcontext():
+            for mb in micros:
+                loss = model(**mb).loss / grad_acc
+                loss.backward()
+            clip_grad_norm_flat_(opt.flat.flat_grad, 1.0)
+            opt.step()
+            opt.zero_grad()
+        if main_stream:
+            torch.cuda.current_stream().wait_stream(main_stream)
 
     if args.advance_steps:
         # tracker-only advance: the inner steps are identical regardless of
