@@ -210,6 +210,17 @@ def train(config: Config):
 
     step = start_step * gradient_accumulation_steps - 1
     max_num_peers = 0
+    # high-priority main stream (same as bench.py): the side-stream dW
+    # GEMMs (ops.py DK_DW_ASYNC) otherwise starve small main-stream kernels
+    # at workgroup-dispatch arbitration.  DK_MAIN_PRIO=0 disables.
+    main_stream = (torch.cuda.Stream(priority=-1)
+                   if device.type == "cuda" and os.environ.get("DK_MAIN_PRIO", "1") != "0"
+                   else None)
+    import contextlib
+
+    def _main_stream_ctx():
+        return torch.cuda.stream(main_stream) if main_stream else contextlib.nullcontext()
+
     for batch in train_dataloader:
         step += 1
         real_step = (step + 1) // gradient_accumulation_steps
@@ -224,10 +235,13 @@ def train(config: Config):
         for key in batch:
             batch[key] = batch[key].to(device)
 
-        outputs = model(**batch)
-        loss = outputs.loss / gradient_accumulation_steps
-        loss_batch += loss.detach()
-        scaler.scale(loss).backward()
+        with _main_stream_ctx():
+            outputs = model(**batch)
+            loss = outputs.loss / gradient_accumulation_steps
+            loss_batch += loss.detach()
+            scaler.scale(loss).backward()
+        if main_stream:
+            torch.cuda.current_stream().wait_stream(main_stream)
 
         if logging_activations_steps:
             for handle in handles:
