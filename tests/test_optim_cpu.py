@@ -115,3 +115,49 @@ def test_clip_flat_matches_torch_clip_cpu():
     assert tn.item() == pytest.approx(tn_ref.item(), rel=1e-6)
     for g, rg in zip(flat.grad_views(), ref_grads):
         assert torch.allclose(g, rg, atol=1e-6)
+
+
+def test_ckpt_topk_gc(tmp_path):
+    """ckpt_utils.py:170-179 semantics: keep the newest topk checkpoint dirs
+    by step number, delete the rest."""
+    from opendiloco_amd.ckpt import delete_old_checkpoints
+
+    for step in (5, 10, 20, 40):
+        (tmp_path / f"model_step_{step}").mkdir()
+    (tmp_path / "not_a_ckpt").mkdir()
+    deleted = delete_old_checkpoints(str(tmp_path), topk=2)
+    assert sorted(int(d.split("_")[-1]) for d in deleted) == [5, 10]
+    left = sorted(p.name for p in tmp_path.iterdir())
+    assert left == ["model_step_20", "model_step_40", "not_a_ckpt"]
+
+
+def test_dw_nchunk_table_and_fallback():
+    """The measured chunk table is hit for the swept shapes; unknown shapes
+    fall back to the coverage heuristic with the >=2048-token floor."""
+    from opendiloco_amd.ops import _DW_NC_TABLE, _dw_nchunk
+
+    for (R, N, K), want in _DW_NC_TABLE.items():
+        assert _dw_nchunk(R, N, K) == want
+    # unknown shape: heuristic, pow2, chunk >= 2048 tokens
+    nc = _dw_nchunk(8192, 512, 512)
+    assert nc & (nc - 1) == 0
+    assert 8192 // nc >= 2048
+    assert _dw_nchunk(2048, 10_000, 10_000) == 1  # floor: cannot split
+
+
+def test_compression_kwargs_surface():
+    """utils.py:83-121 mapping: None/fp16/scaled-fp16/uniform8bit supported,
+    quantile8bit/blockwise8bit documented NotImplementedError, junk rejected."""
+    import pytest as _pytest
+
+    from opendiloco_amd.utils_compat import get_compression_kwargs
+
+    assert get_compression_kwargs(None) == {"grad_compression": None}
+    assert get_compression_kwargs("fp16") == {"grad_compression": "fp16"}
+    assert get_compression_kwargs("scaled-fp16") == {"grad_compression": "fp16"}
+    assert get_compression_kwargs("uniform8bit") == {"grad_compression": "uniform8bit"}
+    for name in ("quantile8bit", "blockwise8bit"):
+        with _pytest.raises(NotImplementedError):
+            get_compression_kwargs(name)
+    with _pytest.raises(ValueError):
+        get_compression_kwargs("nope")
