@@ -585,9 +585,11 @@ __global__ void attn_bwd_pre_kernel(float* __restrict__ delta,
     for (int q4 = lo; q4 < quads; q4 += 16) {
       shortx4 dv = *(const shortx4*)(do_ + base + q4 * 4);
       shortx4 ov = *(const shortx4*)(o + base + q4 * 4);
+      float df[4], of[4];
+      packed_to_f32<DT, 4>(&dv, df);
+      packed_to_f32<DT, 4>(&ov, of);
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        s += TR::toF(((const T*)&dv)[j]) * TR::toF(((const T*)&ov)[j]);
+      for (int j = 0; j < 4; ++j) s += df[j] * of[j];
     }
 #pragma unroll
     for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, DK_WAVE);
